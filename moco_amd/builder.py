@@ -1,0 +1,149 @@
+"""MoCo: Momentum Contrast, MI355X-native implementation.
+
+API and checkpoint-layout parity with the reference ``moco/builder.py``
+(class MoCo: __init__:16, _momentum_update_key_encoder:51,
+_dequeue_and_enqueue:59, _batch_shuffle_ddp:76, _batch_unshuffle_ddp:106,
+forward:126, concat_all_gather:166), re-designed for the hardware:
+
+- EMA of the key encoder is ONE fused multi-tensor HIP launch instead of a
+  161-iteration Python loop.
+- The contrastive logits [q.k+ | q @ queue]/T are one fused MFMA kernel; the
+  queue's GEMM working copy is stored key-per-row ([K, C] bf16) so both the
+  forward and backward GEMM read K-contiguous bf16 fragments.
+- Shuffling-BN derives its permutation from a shared per-step seed (zero
+  communication, replaces the rank-0 broadcast) and exchanges only the rows
+  each peer needs (all-to-all over xGMI instead of a full all-gather).
+- The canonical queue buffer keeps the reference layout/dtype
+  ([dim, K] fp32, key name ``queue``; ``queue_ptr`` [1]) so checkpoints
+  interoperate.
+
+Invariants preserved (reference forward, builder.py:126-163): EMA fires
+before the key forward, once per step; the queue is updated after the logits
+(current keys are positives, the queue holds past batches only); gradients
+flow only through q; the queue is L2-normalized at init and replicated
+identically on every rank.
+"""
+
+from typing import Callable, Optional
+
+import torch
+import torch.nn as nn
+
+from . import parallel
+from .ops import (MultiTensorEMA, contrastive_logits, l2_normalize,
+                  queue_enqueue_, softmax_ce_label0)
+
+
+class MoCo(nn.Module):
+    def __init__(self, base_encoder: Callable[..., nn.Module], dim: int = 128,
+                 K: int = 65536, m: float = 0.999, T: float = 0.07,
+                 mlp: bool = False, shuffle_seed: int = 0):
+        super().__init__()
+        self.K = K
+        self.m = m
+        self.T = T
+        self.shuffle_seed = shuffle_seed
+
+        self.encoder_q = base_encoder(num_classes=dim)
+        self.encoder_k = base_encoder(num_classes=dim)
+
+        if mlp:  # MoCo v2 projection head (reference builder.py:24-28)
+            dim_mlp = self.encoder_q.fc.weight.shape[1]
+            self.encoder_q.fc = nn.Sequential(
+                nn.Linear(dim_mlp, dim_mlp), nn.ReLU(), self.encoder_q.fc)
+            self.encoder_k.fc = nn.Sequential(
+                nn.Linear(dim_mlp, dim_mlp), nn.ReLU(), self.encoder_k.fc)
+
+        for param_q, param_k in zip(self.encoder_q.parameters(),
+                                    self.encoder_k.parameters()):
+            param_k.data.copy_(param_q.data)
+            param_k.requires_grad = False
+
+        # canonical queue: [dim, K] fp32, L2-normalized columns
+        # (reference builder.py:36-38)
+        self.register_buffer("queue",
+                             nn.functional.normalize(torch.randn(dim, K),
+                                                     dim=0))
+        self.register_buffer("queue_ptr", torch.zeros(1, dtype=torch.long))
+
+        self._ema: Optional[MultiTensorEMA] = None
+        self._queue_kc: Optional[torch.Tensor] = None  # [K, dim] bf16 (GPU)
+        self._step = 0
+
+    # -- key-encoder EMA ----------------------------------------------------
+    @torch.no_grad()
+    def _momentum_update_key_encoder(self):
+        if self._ema is None:
+            self._ema = MultiTensorEMA(list(self.encoder_k.parameters()),
+                                       list(self.encoder_q.parameters()))
+        self._ema.step(self.m)
+
+    # -- queue --------------------------------------------------------------
+    def _working_queue(self) -> torch.Tensor:
+        """The [K, C] layout the GEMM kernels consume.
+
+        On GPU this is a bf16 copy maintained incrementally by the enqueue
+        kernel; on CPU it is just a transposed view of the fp32 queue.
+        """
+        if self.queue.is_cuda:
+            if self._queue_kc is None or \
+                    self._queue_kc.device != self.queue.device:
+                self._queue_kc = self.queue.t().contiguous().to(torch.bfloat16)
+            return self._queue_kc
+        return self.queue.t()
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        self._queue_kc = None  # rebuilt lazily from the loaded fp32 queue
+        self._ema = None
+        super()._load_from_state_dict(*args, **kwargs)
+
+    @torch.no_grad()
+    def _dequeue_and_enqueue(self, keys: torch.Tensor):
+        keys = parallel.concat_all_gather(keys)  # C6: all ranks enqueue all
+        qkc = self._queue_kc if self.queue.is_cuda else None
+        queue_enqueue_(self.queue, qkc, self.queue_ptr, keys)
+
+    # -- shuffling-BN (world-aware; see parallel/comm.py) -------------------
+    @torch.no_grad()
+    def _batch_shuffle(self, x: torch.Tensor, step: int):
+        return parallel.shuffle_exchange(x, step, self.shuffle_seed)
+
+    @torch.no_grad()
+    def _batch_unshuffle(self, x: torch.Tensor, idx_unshuffle: torch.Tensor):
+        return parallel.unshuffle_exchange(x, idx_unshuffle)
+
+    # -- forward ------------------------------------------------------------
+    def forward(self, im_q: torch.Tensor, im_k: torch.Tensor,
+                step: Optional[int] = None):
+        """Compute contrastive logits and labels.
+
+        Returns (logits [N, 1+K], labels [N] all-zero) — parity with the
+        reference forward (builder.py:126-163).
+        """
+        if step is None:
+            step = self._step
+            self._step += 1
+
+        q = self.encoder_q(im_q)
+        q = l2_normalize(q, dim=1)
+
+        with torch.no_grad():
+            self._momentum_update_key_encoder()
+            im_k_sh, idx_unshuffle = self._batch_shuffle(im_k, step)
+            k = self.encoder_k(im_k_sh)
+            k = l2_normalize(k, dim=1)
+            k = self._batch_unshuffle(k, idx_unshuffle)
+
+        if q.is_cuda:
+            q = q.to(torch.bfloat16)
+            k = k.to(torch.bfloat16)
+        logits = contrastive_logits(q, k, self._working_queue(), self.T)
+        labels = torch.zeros(logits.shape[0], dtype=torch.long,
+                             device=logits.device)
+        self._dequeue_and_enqueue(k)
+        return logits, labels
+
+
+def moco_loss(logits: torch.Tensor) -> torch.Tensor:
+    """InfoNCE with the positive at index 0 — fused softmax-CE."""
+    return softmax_ce_label0(logits)

@@ -1,0 +1,13 @@
+from .resnet import (MODEL_REGISTRY, ResNet, build_model, resnet18, resnet34,
+                     resnet50, resnet101, resnet152)
+
+__all__ = [
+    "ResNet",
+    "MODEL_REGISTRY",
+    "build_model",
+    "resnet18",
+    "resnet34",
+    "resnet50",
+    "resnet101",
+    "resnet152",
+]

@@ -1,0 +1,13 @@
+from .meters import AverageMeter, ProgressMeter, accuracy
+from .checkpoint import save_checkpoint, load_pretrained_backbone, sanity_check
+from .lr import adjust_learning_rate
+
+__all__ = [
+    "AverageMeter",
+    "ProgressMeter",
+    "accuracy",
+    "save_checkpoint",
+    "load_pretrained_backbone",
+    "sanity_check",
+    "adjust_learning_rate",
+]
