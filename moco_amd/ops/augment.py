@@ -245,12 +245,13 @@ def _apply_blur(x: torch.Tensor, params: torch.Tensor) -> torch.Tensor:
         k = torch.exp(-0.5 * (t / sig) ** 2)
         k = k / k.sum()
         img = x[i:i + 1]
+        # clamp-to-edge padding (matches the HIP kernel's edge handling)
+        img = torch.nn.functional.pad(img, (R, R, 0, 0), mode="replicate")
         img = torch.nn.functional.conv2d(
-            img, k.view(1, 1, 1, -1).expand(3, 1, 1, -1), padding=(0, R),
-            groups=3)
+            img, k.view(1, 1, 1, -1).expand(3, 1, 1, -1), groups=3)
+        img = torch.nn.functional.pad(img, (0, 0, R, R), mode="replicate")
         img = torch.nn.functional.conv2d(
-            img, k.view(1, 1, -1, 1).expand(3, 1, -1, 1), padding=(R, 0),
-            groups=3)
+            img, k.view(1, 1, -1, 1).expand(3, 1, -1, 1), groups=3)
         out[i] = img[0]
     return out
 

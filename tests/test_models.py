@@ -1,0 +1,41 @@
+import torch
+
+from moco_amd.models.resnet import resnet18, resnet50
+
+
+def test_resnet50_param_count():
+    # standard ResNet-50 with a 1000-way fc: 25,557,032 parameters
+    m = resnet50(num_classes=1000)
+    assert sum(p.numel() for p in m.parameters()) == 25_557_032
+
+
+def test_resnet18_param_count():
+    m = resnet18(num_classes=1000)
+    assert sum(p.numel() for p in m.parameters()) == 11_689_512
+
+
+def test_resnet_forward_shapes():
+    m = resnet18(num_classes=128)
+    for hw in (32, 64):
+        out = m(torch.randn(2, 3, hw, hw))
+        assert out.shape == (2, 128)
+
+
+def test_resnet50_forward_and_grad():
+    m = resnet50(num_classes=16)
+    out = m(torch.randn(2, 3, 64, 64))
+    out.sum().backward()
+    assert m.conv1.weight.grad is not None
+    assert m.fc.weight.grad is not None
+
+
+def test_state_dict_key_layout():
+    # the checkpoint namespace is a compatibility contract
+    keys = set(resnet50(num_classes=10).state_dict().keys())
+    for expected in [
+        "conv1.weight", "bn1.weight", "bn1.running_mean",
+        "layer1.0.conv1.weight", "layer1.0.downsample.0.weight",
+        "layer1.0.downsample.1.running_var", "layer4.2.bn3.bias",
+        "fc.weight", "fc.bias",
+    ]:
+        assert expected in keys

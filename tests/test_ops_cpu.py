@@ -1,0 +1,108 @@
+"""CPU-path numerics for the ops layer (these same implementations are the
+oracle the GPU kernel tests compare against)."""
+
+import math
+
+import torch
+
+from moco_amd.ops import (contrastive_logits, l2_normalize, queue_enqueue_,
+                          softmax_ce_label0)
+from moco_amd.ops.multi_tensor import FusedSGD, MultiTensorEMA
+
+
+def test_l2_normalize_matches_torch():
+    x = torch.randn(32, 128, requires_grad=True)
+    y = l2_normalize(x, dim=1)
+    ref = torch.nn.functional.normalize(x, dim=1)
+    assert torch.allclose(y, ref, atol=1e-6)
+    g = torch.randn_like(y)
+    y.backward(g)
+    x2 = x.detach().clone().requires_grad_(True)
+    torch.nn.functional.normalize(x2, dim=1).backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+
+
+def test_contrastive_logits_matches_einsum():
+    N, C, K = 8, 128, 64
+    T = 0.07
+    q = torch.randn(N, C, requires_grad=True)
+    k = torch.randn(N, C)
+    queue_kc = torch.randn(K, C)
+    logits = contrastive_logits(q, k, queue_kc, T)
+    l_pos = torch.einsum("nc,nc->n", q, k).unsqueeze(-1)
+    l_neg = torch.einsum("nc,ck->nk", q, queue_kc.t())
+    ref = torch.cat([l_pos, l_neg], dim=1) / T
+    assert torch.allclose(logits, ref, atol=1e-5)
+
+    g = torch.randn_like(logits)
+    logits.backward(g)
+    q2 = q.detach().clone().requires_grad_(True)
+    l_pos2 = torch.einsum("nc,nc->n", q2, k).unsqueeze(-1)
+    l_neg2 = torch.einsum("nc,ck->nk", q2, queue_kc.t())
+    (torch.cat([l_pos2, l_neg2], dim=1) / T).backward(g)
+    assert torch.allclose(q.grad, q2.grad, atol=1e-4)
+
+
+def test_softmax_ce_label0_matches_cross_entropy():
+    N, K1 = 16, 257
+    logits = torch.randn(N, K1, requires_grad=True)
+    loss = softmax_ce_label0(logits)
+    logits2 = logits.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(
+        logits2, torch.zeros(N, dtype=torch.long))
+    assert torch.allclose(loss, ref, atol=1e-6)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(logits.grad, logits2.grad, atol=1e-6)
+
+
+def test_queue_enqueue_ring():
+    C, K, B = 16, 32, 8
+    queue = torch.zeros(C, K)
+    ptr = torch.zeros(1, dtype=torch.long)
+    batches = [torch.randn(B, C) for _ in range(5)]
+    for i, keys in enumerate(batches):
+        queue_enqueue_(queue, None, ptr, keys)
+        assert ptr.item() == ((i + 1) * B) % K
+    # after 5 batches of 8 into K=32: columns hold batches [4,1,2,3] rotated
+    assert torch.allclose(queue[:, 0:8], batches[4].t())
+    assert torch.allclose(queue[:, 8:16], batches[1].t())
+    assert torch.allclose(queue[:, 24:32], batches[3].t())
+
+
+def test_multi_tensor_ema_matches_loop():
+    torch.manual_seed(0)
+    dst = [torch.randn(7), torch.randn(3, 5), torch.randn(130000)]
+    src = [torch.randn_like(t) for t in dst]
+    expect = [0.99 * d + 0.01 * s for d, s in zip(dst, src)]
+    MultiTensorEMA(dst, src).step(0.99)
+    for d, e in zip(dst, expect):
+        assert torch.allclose(d, e, atol=1e-6)
+
+
+def test_fused_sgd_matches_torch_sgd():
+    torch.manual_seed(0)
+    p1 = torch.randn(100, requires_grad=True)
+    p2 = p1.detach().clone().requires_grad_(True)
+    opt1 = FusedSGD([p1], lr=0.1, momentum=0.9, weight_decay=1e-4)
+    opt2 = torch.optim.SGD([p2], lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for _ in range(5):
+        g = torch.randn(100)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        opt1.step()
+        opt2.step()
+    assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_fused_sgd_state_roundtrip():
+    p = torch.randn(10, requires_grad=True)
+    opt = FusedSGD([p], lr=0.1, momentum=0.9)
+    p.grad = torch.randn(10)
+    opt.step()
+    state = opt.state_dict()
+    opt2 = FusedSGD([p], lr=0.1, momentum=0.9)
+    opt2.load_state_dict(state)
+    b1 = opt.state[p]["momentum_buffer"]
+    b2 = opt2.state[p]["momentum_buffer"]
+    assert torch.allclose(b1, b2)
