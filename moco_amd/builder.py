@@ -137,7 +137,8 @@ class MoCo(nn.Module):
         if q.is_cuda:
             q = q.to(torch.bfloat16)
             k = k.to(torch.bfloat16)
-        logits = contrastive_logits(q, k, self._working_queue(), self.T)
+        logits = contrastive_logits(q, k, self._working_queue(), self.queue,
+                                    self.T)
         labels = torch.zeros(logits.shape[0], dtype=torch.long,
                              device=logits.device)
         self._dequeue_and_enqueue(k)

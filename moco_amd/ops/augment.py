@@ -213,9 +213,13 @@ def _apply_jitter(x: torch.Tensor, params: torch.Tensor) -> torch.Tensor:
     co = params[:, P_CO, None, None, None]
     sa = params[:, P_SA, None, None, None]
     hu = params[:, P_HU]
+    # contrast blends toward the crop's gray mean; like the HIP kernel we use
+    # the PRE-brightness mean scaled by the brightness factor (exact for
+    # unclipped pixels) so the single gray-mean reduction of pass 1 suffices
+    gray_pre = (GRAY_R * x[:, 0] + GRAY_G * x[:, 1] + GRAY_B * x[:, 2])
+    mean_g = (gray_pre.mean(dim=(1, 2))[:, None, None, None]
+              * br).clamp(0, 1)
     x = (x * br).clamp(0, 1)
-    gray = (GRAY_R * x[:, 0] + GRAY_G * x[:, 1] + GRAY_B * x[:, 2])
-    mean_g = gray.mean(dim=(1, 2))[:, None, None, None]
     x = (co * x + (1 - co) * mean_g).clamp(0, 1)
     gray = (GRAY_R * x[:, 0] + GRAY_G * x[:, 1] + GRAY_B * x[:, 2])[:, None]
     x = (sa * x + (1 - sa) * gray).clamp(0, 1)

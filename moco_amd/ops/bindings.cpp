@@ -1,0 +1,222 @@
+// Python <-> HIP bindings for moco_amd._C.
+// This is the only torch-dependent TU; the kernels live in hip/*.hip behind
+// the C ABI of hip/launchers.h.
+#include <torch/extension.h>
+
+#include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+
+#include "hip/launchers.h"
+
+namespace {
+
+void* cur_stream() {
+  return (void*)at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define CHECK_LAUNCH(expr)                                            \
+  do {                                                                \
+    int rc_ = (expr);                                                 \
+    TORCH_CHECK(rc_ == 0, #expr " failed with hipError_t=", rc_);     \
+  } while (0)
+
+#define CHECK_CUDA_CONTIG(t)                                          \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t,               \
+              " must be a contiguous GPU tensor")
+
+bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> l2norm_rows_fwd(torch::Tensor x) {
+  CHECK_CUDA_CONTIG(x);
+  TORCH_CHECK(x.dim() == 2, "x must be [N, C]");
+  TORCH_CHECK(is_bf16(x) || x.scalar_type() == torch::kFloat32);
+  at::cuda::CUDAGuard guard(x.device());
+  auto y = torch::empty_like(x);
+  auto inv_norm = torch::empty({x.size(0)},
+                               x.options().dtype(torch::kFloat32));
+  CHECK_LAUNCH(launch_l2norm_fwd(cur_stream(), x.data_ptr(), y.data_ptr(),
+                                 inv_norm.data_ptr<float>(), x.size(0),
+                                 x.size(1), is_bf16(x)));
+  return {y, inv_norm};
+}
+
+torch::Tensor l2norm_rows_bwd(torch::Tensor g, torch::Tensor y,
+                              torch::Tensor inv_norm) {
+  CHECK_CUDA_CONTIG(g);
+  CHECK_CUDA_CONTIG(y);
+  TORCH_CHECK(g.sizes() == y.sizes() && g.scalar_type() == y.scalar_type());
+  at::cuda::CUDAGuard guard(g.device());
+  auto gx = torch::empty_like(g);
+  CHECK_LAUNCH(launch_l2norm_bwd(cur_stream(), g.data_ptr(), y.data_ptr(),
+                                 inv_norm.data_ptr<float>(), gx.data_ptr(),
+                                 g.size(0), g.size(1), is_bf16(g)));
+  return gx;
+}
+
+// ---------------------------------------------------------------------------
+torch::Tensor contrastive_fwd(torch::Tensor q, torch::Tensor k,
+                              torch::Tensor queue_kc, double invT) {
+  CHECK_CUDA_CONTIG(q);
+  CHECK_CUDA_CONTIG(k);
+  CHECK_CUDA_CONTIG(queue_kc);
+  TORCH_CHECK(is_bf16(q) && is_bf16(k) && is_bf16(queue_kc),
+              "contrastive_fwd expects bf16 q/k/queue_kc");
+  const int64_t N = q.size(0), C = q.size(1), K = queue_kc.size(0);
+  TORCH_CHECK(k.size(0) == N && k.size(1) == C && queue_kc.size(1) == C);
+  at::cuda::CUDAGuard guard(q.device());
+  auto out = torch::empty({N, 1 + K}, q.options().dtype(torch::kFloat32));
+  CHECK_LAUNCH(launch_contrastive_fwd(cur_stream(), q.data_ptr(),
+                                      k.data_ptr(), queue_kc.data_ptr(),
+                                      out.data_ptr<float>(), N, C, K,
+                                      (float)invT));
+  return out;
+}
+
+torch::Tensor contrastive_bwd(torch::Tensor g, torch::Tensor k,
+                              torch::Tensor queue_ck, double invT) {
+  CHECK_CUDA_CONTIG(g);
+  CHECK_CUDA_CONTIG(k);
+  CHECK_CUDA_CONTIG(queue_ck);
+  TORCH_CHECK(g.scalar_type() == torch::kFloat32 &&
+              queue_ck.scalar_type() == torch::kFloat32 && is_bf16(k));
+  const int64_t N = g.size(0), C = queue_ck.size(0), K = queue_ck.size(1);
+  TORCH_CHECK(g.size(1) == 1 + K && k.size(0) == N && k.size(1) == C);
+  at::cuda::CUDAGuard guard(g.device());
+  auto dq = torch::zeros({N, C}, g.options());
+  CHECK_LAUNCH(launch_contrastive_bwd(cur_stream(), g.data_ptr<float>(),
+                                      k.data_ptr(),
+                                      queue_ck.data_ptr<float>(),
+                                      dq.data_ptr<float>(), N, C, K,
+                                      (float)invT));
+  return dq;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> softmax_ce0_fwd(torch::Tensor logits) {
+  CHECK_CUDA_CONTIG(logits);
+  TORCH_CHECK(logits.scalar_type() == torch::kFloat32 && logits.dim() == 2);
+  at::cuda::CUDAGuard guard(logits.device());
+  const int64_t N = logits.size(0), M = logits.size(1);
+  auto opts = logits.options();
+  auto loss_rows = torch::empty({N}, opts);
+  auto row_max = torch::empty({N}, opts);
+  auto row_lse = torch::empty({N}, opts);
+  CHECK_LAUNCH(launch_softmax_ce0_fwd(
+      cur_stream(), logits.data_ptr<float>(), loss_rows.data_ptr<float>(),
+      row_max.data_ptr<float>(), row_lse.data_ptr<float>(), N, M));
+  return {loss_rows, row_max, row_lse};
+}
+
+torch::Tensor softmax_ce0_bwd(torch::Tensor logits, torch::Tensor row_max,
+                              torch::Tensor row_lse, double scale) {
+  CHECK_CUDA_CONTIG(logits);
+  at::cuda::CUDAGuard guard(logits.device());
+  auto dlogits = torch::empty_like(logits);
+  CHECK_LAUNCH(launch_softmax_ce0_bwd(
+      cur_stream(), logits.data_ptr<float>(), row_max.data_ptr<float>(),
+      row_lse.data_ptr<float>(), dlogits.data_ptr<float>(), (float)scale,
+      logits.size(0), logits.size(1)));
+  return dlogits;
+}
+
+// ---------------------------------------------------------------------------
+void queue_enqueue(torch::Tensor queue, c10::optional<torch::Tensor> queue_kc,
+                   torch::Tensor keys, int64_t ptr) {
+  CHECK_CUDA_CONTIG(queue);
+  CHECK_CUDA_CONTIG(keys);
+  TORCH_CHECK(queue.scalar_type() == torch::kFloat32);
+  const int64_t C = queue.size(0), K = queue.size(1), B = keys.size(0);
+  TORCH_CHECK(keys.size(1) == C && ptr + B <= K);
+  void* kc = nullptr;
+  if (queue_kc.has_value()) {
+    CHECK_CUDA_CONTIG(*queue_kc);
+    TORCH_CHECK(is_bf16(*queue_kc) && queue_kc->size(0) == K &&
+                queue_kc->size(1) == C);
+    kc = queue_kc->data_ptr();
+  }
+  at::cuda::CUDAGuard guard(queue.device());
+  CHECK_LAUNCH(launch_queue_enqueue(cur_stream(),
+                                    queue.data_ptr<float>(), kc,
+                                    keys.data_ptr(), is_bf16(keys), C, K, B,
+                                    ptr));
+}
+
+// ---------------------------------------------------------------------------
+void multi_tensor_ema(torch::Tensor pd, torch::Tensor ps, torch::Tensor sizes,
+                      double m) {
+  CHECK_CUDA_CONTIG(pd);
+  CHECK_CUDA_CONTIG(ps);
+  CHECK_CUDA_CONTIG(sizes);
+  at::cuda::CUDAGuard guard(pd.device());
+  CHECK_LAUNCH(launch_multi_tensor_ema(
+      cur_stream(), pd.data_ptr<int64_t>(), ps.data_ptr<int64_t>(),
+      sizes.data_ptr<int32_t>(), pd.size(0), (float)m));
+}
+
+void multi_tensor_sgd(torch::Tensor pp, torch::Tensor pg, torch::Tensor pb,
+                      torch::Tensor sizes, double lr, double momentum,
+                      double wd) {
+  at::cuda::CUDAGuard guard(pp.device());
+  CHECK_LAUNCH(launch_multi_tensor_sgd(
+      cur_stream(), pp.data_ptr<int64_t>(), pg.data_ptr<int64_t>(),
+      pb.data_ptr<int64_t>(), sizes.data_ptr<int32_t>(), pp.size(0),
+      (float)lr, (float)momentum, (float)wd));
+}
+
+void ema_flat(torch::Tensor dst, torch::Tensor src, double m) {
+  CHECK_CUDA_CONTIG(dst);
+  CHECK_CUDA_CONTIG(src);
+  TORCH_CHECK(dst.scalar_type() == torch::kFloat32 &&
+              src.scalar_type() == torch::kFloat32 &&
+              dst.numel() == src.numel());
+  at::cuda::CUDAGuard guard(dst.device());
+  CHECK_LAUNCH(launch_ema_flat(cur_stream(), dst.data_ptr<float>(),
+                               src.data_ptr<float>(), dst.numel(),
+                               (float)m));
+}
+
+// ---------------------------------------------------------------------------
+torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
+                               int64_t S, bool out_bf16, bool channels_last) {
+  CHECK_CUDA_CONTIG(src);
+  CHECK_CUDA_CONTIG(params);
+  TORCH_CHECK(src.scalar_type() == torch::kFloat32 && src.dim() == 4 &&
+              src.size(1) == 3, "src must be fp32 [B,3,H,W]");
+  TORCH_CHECK(params.scalar_type() == torch::kFloat32 &&
+              params.size(1) == 12);
+  const int64_t B = src.size(0), H = src.size(2), W = src.size(3);
+  const int64_t M = params.size(0);
+  at::cuda::CUDAGuard guard(src.device());
+  auto f32 = src.options();
+  auto out_opts = out_bf16 ? f32.dtype(torch::kBFloat16) : f32;
+  torch::Tensor out = torch::empty({M, 3, S, S}, out_opts);
+  if (channels_last)
+    out = out.contiguous(torch::MemoryFormat::ChannelsLast);
+  auto tmp1 = torch::empty({M, 3, S, S}, f32);
+  auto tmp2 = torch::empty({M, 3, S, S}, f32);
+  auto graysum = torch::zeros({M}, f32);
+  CHECK_LAUNCH(launch_two_crop_augment(
+      cur_stream(), src.data_ptr<float>(), params.data_ptr<float>(),
+      out.data_ptr(), tmp1.data_ptr<float>(), tmp2.data_ptr<float>(),
+      graysum.data_ptr<float>(), B, M, H, W, S, out_bf16, channels_last));
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("l2norm_rows_fwd", &l2norm_rows_fwd);
+  mod.def("l2norm_rows_bwd", &l2norm_rows_bwd);
+  mod.def("contrastive_fwd", &contrastive_fwd);
+  mod.def("contrastive_bwd", &contrastive_bwd);
+  mod.def("softmax_ce0_fwd", &softmax_ce0_fwd);
+  mod.def("softmax_ce0_bwd", &softmax_ce0_bwd);
+  mod.def("queue_enqueue", &queue_enqueue);
+  mod.def("multi_tensor_ema", &multi_tensor_ema);
+  mod.def("multi_tensor_sgd", &multi_tensor_sgd);
+  mod.def("ema_flat", &ema_flat);
+  mod.def("two_crop_augment", &two_crop_augment);
+}

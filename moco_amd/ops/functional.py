@@ -63,21 +63,29 @@ def l2_normalize(x: torch.Tensor, dim: int = 1) -> torch.Tensor:
 
 class _ContrastiveLogits(torch.autograd.Function):
     """logits[n, 0]   = (q[n] . k[n]) / T
-       logits[n, 1+j] = (q[n] . queue_kc[j]) / T
+       logits[n, 1+j] = (q[n] . queue[:, j]) / T
 
     Gradients flow to q only (k and queue are detached by construction:
-    reference builder.py:133-139,149).  queue_kc is the [K, C] row-major
-    (key-per-row) working layout — chosen so both the forward NT-GEMM and
-    the backward NT-GEMM read K-contiguous fragments (see ops/hip/contrastive.hip).
+    reference builder.py:133-139,149).
+
+    Layouts (MFMA NT-GEMMs want the CONTRACTION index contiguous in both
+    operands — see ops/hip/contrastive.hip):
+    - forward contracts over C: uses queue_kc [K, C] (key-per-row, bf16 on
+      GPU) so q rows and queue rows both read C-contiguous fragments;
+    - backward contracts over K: uses queue_ck [C, K] (the canonical fp32
+      ring buffer) so grad rows and queue_ck rows both read K-contiguous
+      fragments.
     """
 
     @staticmethod
     def forward(ctx, q: torch.Tensor, k: torch.Tensor,
-                queue_kc: torch.Tensor, T: float) -> torch.Tensor:
+                queue_kc: torch.Tensor, queue_ck: torch.Tensor,
+                T: float) -> torch.Tensor:
         q = q.contiguous()
         k = k.contiguous()
         if use_hip(q):
-            logits = ext().contrastive_fwd(q, k, queue_kc, 1.0 / T)
+            logits = ext().contrastive_fwd(q, k, queue_kc.contiguous(),
+                                           1.0 / T)
         else:
             qf, kf, Qf = q.float(), k.float(), queue_kc.float()
             l_pos = (qf * kf).sum(dim=1, keepdim=True)
@@ -86,26 +94,28 @@ class _ContrastiveLogits(torch.autograd.Function):
         # snapshot: the enqueue that follows this forward overwrites B rows of
         # the ring buffer in place (reference dodges this with
         # queue.clone().detach() at builder.py:149)
-        ctx.save_for_backward(k, queue_kc.clone())
+        ctx.save_for_backward(k, queue_ck.clone())
         ctx.inv_T = 1.0 / T
         ctx.q_dtype = q.dtype
         return logits
 
     @staticmethod
     def backward(ctx, grad_logits: torch.Tensor):
-        k, queue_kc = ctx.saved_tensors
+        k, queue_ck = ctx.saved_tensors
         if use_hip(grad_logits):
-            gq = ext().contrastive_bwd(grad_logits.contiguous(), k, queue_kc,
-                                       ctx.inv_T)
+            gq = ext().contrastive_bwd(grad_logits.contiguous(), k,
+                                       queue_ck.contiguous(), ctx.inv_T)
         else:
             g = grad_logits.float()
-            gq = (g[:, :1] * k.float() + g[:, 1:] @ queue_kc.float()) * ctx.inv_T
-        return gq.to(ctx.q_dtype), None, None, None
+            gq = (g[:, :1] * k.float() +
+                  g[:, 1:] @ queue_ck.float().t()) * ctx.inv_T
+        return gq.to(ctx.q_dtype), None, None, None, None
 
 
 def contrastive_logits(q: torch.Tensor, k: torch.Tensor,
-                       queue_kc: torch.Tensor, T: float) -> torch.Tensor:
-    return _ContrastiveLogits.apply(q, k, queue_kc, T)
+                       queue_kc: torch.Tensor, queue_ck: torch.Tensor,
+                       T: float) -> torch.Tensor:
+    return _ContrastiveLogits.apply(q, k, queue_kc, queue_ck, T)
 
 
 # ---------------------------------------------------------------------------

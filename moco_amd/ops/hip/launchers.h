@@ -1,0 +1,64 @@
+// C ABI between the torch binding TU (bindings.cpp) and the pure-HIP kernel
+// TUs (*.hip).  `stream` is a hipStream_t passed as void*.
+// Every launcher returns a hipError_t as int (0 == success).
+#pragma once
+
+#include <cstdint>
+
+extern "C" {
+
+// ---- l2norm.hip -----------------------------------------------------------
+// x: [N, C] (bf16 if is_bf16 else fp32) -> y same dtype, inv_norm fp32 [N]
+int launch_l2norm_fwd(void* stream, const void* x, void* y, float* inv_norm,
+                      int64_t N, int64_t C, bool is_bf16);
+int launch_l2norm_bwd(void* stream, const void* g, const void* y,
+                      const float* inv_norm, void* gx, int64_t N, int64_t C,
+                      bool is_bf16);
+
+// ---- contrastive.hip ------------------------------------------------------
+// q,k: [N,C] bf16; queue_kc: [K,C] bf16; out: [N,1+K] fp32
+int launch_contrastive_fwd(void* stream, const void* q, const void* k,
+                           const void* queue_kc, float* out, int64_t N,
+                           int64_t C, int64_t K, float invT);
+// g: [N,1+K] fp32; k: [N,C] bf16; queue_ck: [C,K] fp32; dq (pre-zeroed): [N,C] fp32
+int launch_contrastive_bwd(void* stream, const float* g, const void* k,
+                           const float* queue_ck, float* dq, int64_t N,
+                           int64_t C, int64_t K, float invT);
+
+// ---- softmax_ce.hip -------------------------------------------------------
+// logits: [N, M] fp32
+int launch_softmax_ce0_fwd(void* stream, const float* logits, float* loss_rows,
+                           float* row_max, float* row_lse, int64_t N,
+                           int64_t M);
+int launch_softmax_ce0_bwd(void* stream, const float* logits,
+                           const float* row_max, const float* row_lse,
+                           float* dlogits, float scale, int64_t N, int64_t M);
+
+// ---- queue.hip ------------------------------------------------------------
+// queue: [C,K] fp32; queue_kc: [K,C] bf16 (nullable); keys: [B,C]
+int launch_queue_enqueue(void* stream, float* queue, void* queue_kc,
+                         const void* keys, bool keys_bf16, int64_t C,
+                         int64_t K, int64_t B, int64_t ptr);
+
+// ---- multi_tensor.hip -----------------------------------------------------
+int launch_multi_tensor_ema(void* stream, const int64_t* ptrs_dst,
+                            const int64_t* ptrs_src, const int32_t* sizes,
+                            int64_t nchunks, float m);
+int launch_multi_tensor_sgd(void* stream, const int64_t* ptrs_p,
+                            const int64_t* ptrs_g, const int64_t* ptrs_b,
+                            const int32_t* sizes, int64_t nchunks, float lr,
+                            float momentum, float wd);
+int launch_ema_flat(void* stream, float* dst, const float* src, int64_t n,
+                    float m);
+
+// ---- augment.hip ----------------------------------------------------------
+// src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,
+// NCHW or NHWC per flags); tmp1/tmp2: [M,3,S,S] fp32 workspaces;
+// graysum: [M] fp32 (pre-zeroed)
+int launch_two_crop_augment(void* stream, const float* src,
+                            const float* params, void* out, float* tmp1,
+                            float* tmp2, float* graysum, int64_t B, int64_t M,
+                            int64_t H, int64_t W, int64_t S, bool out_bf16,
+                            bool channels_last);
+
+}  // extern "C"

@@ -28,7 +28,7 @@ def test_contrastive_logits_matches_einsum():
     q = torch.randn(N, C, requires_grad=True)
     k = torch.randn(N, C)
     queue_kc = torch.randn(K, C)
-    logits = contrastive_logits(q, k, queue_kc, T)
+    logits = contrastive_logits(q, k, queue_kc, queue_kc.t().contiguous(), T)
     l_pos = torch.einsum("nc,nc->n", q, k).unsqueeze(-1)
     l_neg = torch.einsum("nc,ck->nk", q, queue_kc.t())
     ref = torch.cat([l_pos, l_neg], dim=1) / T

@@ -1,0 +1,281 @@
+"""GPU kernel numerics: every HIP kernel vs the plain-PyTorch fp32 oracle
+(the same implementations that run on the CPU path).  All tests @gpu."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def to_gpu(*ts, dtype=None):
+    out = []
+    for t in ts:
+        t = t.cuda()
+        if dtype is not None:
+            t = t.to(dtype)
+        out.append(t)
+    return out[0] if len(out) == 1 else out
+
+
+# ---------------------------------------------------------------------------
+# l2norm
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("shape", [(32, 128), (7, 128), (256, 128), (4, 64)])
+def test_l2norm_fwd_bwd(dtype, shape):
+    from moco_amd import _C
+    torch.manual_seed(0)
+    x = torch.randn(*shape)
+    xg = x.cuda().to(dtype)
+    y, inv = _C.l2norm_rows_fwd(xg)
+    ref = torch.nn.functional.normalize(x.float(), dim=1)
+    atol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(y.float().cpu(), ref, atol=atol)
+    assert torch.allclose(inv.cpu(), 1.0 / x.float().norm(dim=1), rtol=1e-4)
+
+    g = torch.randn(*shape)
+    gg = g.cuda().to(dtype)
+    gx = _C.l2norm_rows_bwd(gg, y, inv)
+    x2 = x.clone().requires_grad_(True)
+    torch.nn.functional.normalize(x2, dim=1).backward(g)
+    assert torch.allclose(gx.float().cpu(), x2.grad, atol=5e-2
+                          if dtype == torch.bfloat16 else 1e-5)
+
+
+# ---------------------------------------------------------------------------
+# contrastive logits (MFMA layout check: random asymmetric data)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("N,C,K", [(32, 128, 1024), (8, 128, 64),
+                                   (256, 128, 65536), (7, 128, 256),
+                                   (16, 64, 512), (33, 256, 4096)])
+def test_contrastive_fwd(N, C, K):
+    from moco_amd import _C
+    torch.manual_seed(1)
+    invT = 1.0 / 0.2
+    # unit-norm rows, like real usage
+    q = torch.nn.functional.normalize(torch.randn(N, C), dim=1)
+    k = torch.nn.functional.normalize(torch.randn(N, C), dim=1)
+    queue_kc = torch.nn.functional.normalize(torch.randn(K, C), dim=1)
+    qg, kg, Qg = to_gpu(q, k, queue_kc, dtype=torch.bfloat16)
+    out = _C.contrastive_fwd(qg, kg, Qg, invT)
+    assert out.shape == (N, 1 + K) and out.dtype == torch.float32
+    l_pos = (q * k).sum(1, keepdim=True)
+    l_neg = q @ queue_kc.t()
+    ref = torch.cat([l_pos, l_neg], 1) * invT
+    # bf16 inputs, fp32 accumulate: |logit| <= 1/T = 5
+    assert torch.allclose(out.cpu(), ref, atol=8e-2), \
+        (out.cpu() - ref).abs().max()
+
+
+def test_contrastive_identity_layout():
+    """A = one-hot rows against asymmetric queue: catches any fragment
+    row/col transpose (guide G9/rule 16)."""
+    from moco_amd import _C
+    N, C, K = 16, 128, 64
+    q = torch.zeros(N, C)
+    for n in range(N):
+        q[n, n + 3] = 1.0  # one-hot at shifted positions
+    queue_kc = torch.arange(K * C, dtype=torch.float32).view(K, C) / (K * C)
+    k = torch.zeros(N, C)
+    qg, kg, Qg = to_gpu(q, k, queue_kc, dtype=torch.bfloat16)
+    out = _C.contrastive_fwd(qg, kg, Qg, 1.0)
+    ref = q @ queue_kc.to(torch.bfloat16).float().t()
+    assert torch.allclose(out[:, 1:].cpu(), ref, atol=1e-3), \
+        (out[:, 1:].cpu() - ref).abs().max()
+
+
+@pytest.mark.parametrize("N,C,K", [(32, 128, 1024), (8, 128, 64),
+                                   (256, 128, 65536), (7, 128, 4096)])
+def test_contrastive_bwd(N, C, K):
+    from moco_amd import _C
+    torch.manual_seed(2)
+    invT = 1.0 / 0.2
+    g = torch.randn(N, 1 + K) / math.sqrt(K)  # CE-grad-like magnitude
+    k = torch.nn.functional.normalize(torch.randn(N, C), dim=1)
+    queue_ck = torch.nn.functional.normalize(torch.randn(C, K), dim=0)
+    gg = g.cuda()
+    kg = k.cuda().to(torch.bfloat16)
+    Qg = queue_ck.cuda()
+    dq = _C.contrastive_bwd(gg, kg, Qg, invT)
+    ref = (g[:, :1] * k + g[:, 1:] @ queue_ck.t()) * invT
+    assert torch.allclose(dq.cpu(), ref, atol=5e-2,
+                          rtol=1e-2), (dq.cpu() - ref).abs().max()
+
+
+# ---------------------------------------------------------------------------
+# softmax CE (label 0)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("N,M", [(32, 1025), (8, 65537), (256, 65537),
+                                 (7, 131073), (5, 13)])
+def test_softmax_ce0(N, M):
+    from moco_amd import _C
+    torch.manual_seed(3)
+    logits = (torch.randn(N, M) * 3).cuda()
+    loss_rows, row_max, row_lse = _C.softmax_ce0_fwd(logits)
+    ref_loss = torch.nn.functional.cross_entropy(
+        logits.cpu(), torch.zeros(N, dtype=torch.long), reduction="none")
+    assert torch.allclose(loss_rows.cpu(), ref_loss, atol=1e-4, rtol=1e-5)
+
+    scale = 1.0 / N
+    d = _C.softmax_ce0_bwd(logits, row_max, row_lse, scale)
+    l2 = logits.cpu().clone().requires_grad_(True)
+    torch.nn.functional.cross_entropy(
+        l2, torch.zeros(N, dtype=torch.long)).backward()
+    assert torch.allclose(d.cpu(), l2.grad, atol=1e-6), \
+        (d.cpu() - l2.grad).abs().max()
+
+
+# ---------------------------------------------------------------------------
+# queue enqueue
+# ---------------------------------------------------------------------------
+
+def test_queue_enqueue_gpu():
+    from moco_amd import _C
+    C, K, B = 128, 256, 32
+    torch.manual_seed(4)
+    queue = torch.randn(C, K).cuda()
+    queue_kc = queue.t().contiguous().to(torch.bfloat16)
+    keys = torch.nn.functional.normalize(torch.randn(B, C), dim=1) \
+        .cuda().to(torch.bfloat16)
+    ref = queue.clone()
+    _C.queue_enqueue(queue, queue_kc, keys, 64)
+    ref[:, 64:64 + B] = keys.float().t()
+    assert torch.allclose(queue, ref)
+    assert torch.allclose(queue_kc[64:64 + B].float(), keys.float())
+    # untouched rows of queue_kc intact
+    assert torch.allclose(queue_kc[:64].float(),
+                          queue[:, :64].t().to(torch.bfloat16).float())
+
+
+# ---------------------------------------------------------------------------
+# multi-tensor EMA / SGD
+# ---------------------------------------------------------------------------
+
+def test_multi_tensor_ema_gpu():
+    from moco_amd.ops.multi_tensor import MultiTensorEMA
+    torch.manual_seed(5)
+    dst = [torch.randn(s).cuda() for s in [(7,), (130000,), (64, 64)]]
+    src = [torch.randn_like(t) for t in dst]
+    expect = [0.999 * d + 0.001 * s for d, s in zip(dst, src)]
+    MultiTensorEMA(dst, src).step(0.999)
+    torch.cuda.synchronize()
+    for d, e in zip(dst, expect):
+        assert torch.allclose(d, e, atol=1e-6)
+
+
+def test_fused_sgd_gpu_matches_torch():
+    from moco_amd.ops.multi_tensor import FusedSGD
+    torch.manual_seed(6)
+    shapes = [(64, 3, 7, 7), (1000,), (130001,)]
+    p1 = [torch.randn(s).cuda().requires_grad_(True) for s in shapes]
+    p2 = [p.detach().clone().requires_grad_(True) for p in p1]
+    o1 = FusedSGD(p1, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    o2 = torch.optim.SGD(p2, lr=0.1, momentum=0.9, weight_decay=1e-4)
+    for _ in range(4):
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad = g.clone()
+            b.grad = g.clone()
+        o1.step()
+        o2.step()
+    torch.cuda.synchronize()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+
+
+# ---------------------------------------------------------------------------
+# augmentation pipeline vs torch oracle
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("aug_plus", [False, True])
+def test_two_crop_augment_vs_oracle(aug_plus):
+    from moco_amd import _C
+    from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
+    torch.manual_seed(7)
+    B, H, W, S = 4, 96, 96, 64
+    x = torch.rand(B, 3, H, W)
+    params = sample_two_crop_params(B, H, W, S, aug_plus)
+    out = _C.two_crop_augment(x.cuda(), params.cuda(), S, False, False)
+    ref = _augment_torch(x, params, S)
+    diff = (out.cpu() - ref).abs()
+    assert diff.max() < 5e-2, diff.max()
+    assert (diff > 1e-2).float().mean() < 2e-3
+
+
+def test_two_crop_augment_bf16_channels_last():
+    from moco_amd import _C
+    from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
+    torch.manual_seed(8)
+    B, H, W, S = 2, 64, 64, 32
+    x = torch.rand(B, 3, H, W)
+    params = sample_two_crop_params(B, H, W, S, True)
+    out = _C.two_crop_augment(x.cuda(), params.cuda(), S, True, True)
+    assert out.dtype == torch.bfloat16
+    assert out.is_contiguous(memory_format=torch.channels_last)
+    ref = _augment_torch(x, params, S)
+    assert (out.float().cpu() - ref).abs().max() < 8e-2
+
+
+# ---------------------------------------------------------------------------
+# end-to-end on GPU
+# ---------------------------------------------------------------------------
+
+def test_moco_train_step_gpu():
+    import moco_amd
+    from moco_amd.engine import MoCoStepper
+    from moco_amd.models.resnet import resnet18
+    from moco_amd.ops import FusedSGD
+
+    torch.manual_seed(0)
+    device = torch.device("cuda", 0)
+    model = moco_amd.MoCo(
+        lambda num_classes: resnet18(num_classes=num_classes),
+        dim=128, K=1024, m=0.999, T=0.2, mlp=True)
+    model = model.to(device).to(memory_format=torch.channels_last)
+    opt = FusedSGD(model.parameters(), lr=0.03, momentum=0.9,
+                   weight_decay=1e-4)
+    stepper = MoCoStepper(model, opt, device, aug_plus=True, out_size=64)
+    for i in range(3):
+        raw = torch.rand(16, 3, 96, 96, device=device)
+        loss, logits, labels = stepper.step_from_raw(raw)
+    torch.cuda.synchronize()
+    lv = float(loss.detach())
+    assert lv == lv and 0 < lv < 100
+    assert model.queue_ptr.item() == 48 % 1024
+
+
+def test_moco_gpu_matches_cpu_reference_step():
+    """One full MoCo forward on GPU (HIP kernels, bf16) vs the identical
+    model on CPU (fp32 oracle): logits must agree to bf16 tolerance."""
+    import moco_amd
+    from moco_amd.models.resnet import resnet18
+
+    torch.manual_seed(0)
+    model_cpu = moco_amd.MoCo(
+        lambda num_classes: resnet18(num_classes=num_classes),
+        dim=128, K=256, m=0.999, T=0.2)
+    model_gpu = moco_amd.MoCo(
+        lambda num_classes: resnet18(num_classes=num_classes),
+        dim=128, K=256, m=0.999, T=0.2)
+    model_gpu.load_state_dict(model_cpu.state_dict())
+    model_gpu = model_gpu.cuda()
+    model_cpu.eval()
+    model_gpu.eval()
+
+    im_q = torch.randn(8, 3, 32, 32)
+    im_k = torch.randn(8, 3, 32, 32)
+    with torch.no_grad():
+        lc, _ = model_cpu(im_q, im_k, step=0)
+        lg, _ = model_gpu(im_q.cuda(), im_k.cuda(), step=0)
+    # bf16 encoder + bf16 GEMM vs fp32: loose tolerance, but correlation
+    # must be essentially perfect
+    lcf = lc.flatten()
+    lgf = lg.cpu().flatten()
+    corr = torch.corrcoef(torch.stack([lcf, lgf]))[0, 1]
+    assert corr > 0.999, corr
+    assert (lcf - lgf).abs().max() < 0.5
