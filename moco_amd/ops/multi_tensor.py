@@ -26,10 +26,21 @@ def _build_chunks(tensors_lists: Sequence[Sequence[torch.Tensor]]):
     n_lists = len(tensors_lists)
     ptrs: List[List[int]] = [[] for _ in range(n_lists)]
     sizes: List[int] = []
+
+    def _dense(t):
+        # channels_last weights are dense too; elementwise kernels only need
+        # a dense buffer with IDENTICAL strides across the parallel lists
+        return (t.is_contiguous()
+                or t.is_contiguous(memory_format=torch.channels_last)
+                or t.is_contiguous(memory_format=torch.channels_last_3d))
+
     for tensors in zip(*tensors_lists):
         numel = tensors[0].numel()
         for t in tensors:
-            assert t.numel() == numel and t.is_contiguous()
+            assert t.numel() == numel and _dense(t), \
+                (t.shape, t.stride(), "non-dense tensor in multi-tensor op")
+            assert t.stride() == tensors[0].stride(), \
+                (t.stride(), tensors[0].stride(), "stride mismatch")
         esizes = [t.element_size() for t in tensors]
         base = [t.data_ptr() for t in tensors]
         off = 0
@@ -119,9 +130,16 @@ class FusedSGD(torch.optim.Optimizer):
                 state = self.state[p]
                 if "momentum_buffer" not in state:
                     state["momentum_buffer"] = torch.zeros_like(
-                        p, memory_format=torch.contiguous_format)
+                        p, memory_format=torch.preserve_format)
+                g = p.grad
+                if g.stride() != p.stride():
+                    if p.dim() == 4 and p.is_contiguous(
+                            memory_format=torch.channels_last):
+                        g = g.contiguous(memory_format=torch.channels_last)
+                    else:
+                        g = g.contiguous()
                 params.append(p)
-                grads.append(p.grad.contiguous())
+                grads.append(g)
                 bufs.append(state["momentum_buffer"])
             if not params:
                 continue

@@ -34,7 +34,10 @@ def test_l2norm_fwd_bwd(dtype, shape):
     ref = torch.nn.functional.normalize(x.float(), dim=1)
     atol = 1e-5 if dtype == torch.float32 else 2e-2
     assert torch.allclose(y.float().cpu(), ref, atol=atol)
-    assert torch.allclose(inv.cpu(), 1.0 / x.float().norm(dim=1), rtol=1e-4)
+    # bf16 inputs round before the norm: ~0.4% relative slack
+    inv_rtol = 1e-4 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(inv.cpu(), 1.0 / x.float().norm(dim=1),
+                          rtol=inv_rtol)
 
     g = torch.randn(*shape)
     gg = g.cuda().to(dtype)
