@@ -17,8 +17,10 @@ __global__ __launch_bounds__(256) void ce0_fwd_kernel(
   const float* lr = logits + row * M;
   const int tid = threadIdx.x;
 
-  // online max/sum over this thread's strided slice (vectorized by 4)
-  float m = -INFINITY;
+  // online max/sum over this thread's strided slice (vectorized by 4).
+  // -FLT_MAX, not -inf: threads with an empty slice must combine as
+  // exp(m - nm) = exp(finite), never (-inf) - (-inf) = nan.
+  float m = -3.0e38f;
   float s = 0.f;
   const int64_t M4 = M & ~3LL;
   for (int64_t i = (int64_t)tid * 4; i < M4; i += 256 * 4) {
