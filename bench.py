@@ -39,6 +39,8 @@ def get_args():
     p.add_argument("--crop-size", type=int, default=224)
     p.add_argument("--src-size", type=int, default=256)
     p.add_argument("--pool-batches", type=int, default=4)
+    p.add_argument("--phase-timing", action="store_true",
+                   help="print per-phase GPU-event ms after the timed run")
     return p.parse_args()
 
 
@@ -97,6 +99,15 @@ def main():
 
     for i in range(args.warmup):
         one_step(i)
+
+    if args.phase_timing and not cpu_fallback:
+        stepper.enable_timing(True)
+        one_step(99)
+        torch.cuda.synchronize()
+        if rank == 0:
+            print("phase_ms:", {k: round(v, 3)
+                                for k, v in stepper.phase_times_ms().items()})
+        stepper.enable_timing(False)
 
     if distributed:
         dist.barrier()

@@ -59,17 +59,47 @@ class MoCoStepper:
         """One optimization step on pre-augmented crops.
         Returns (loss, logits, labels)."""
         autocast_on = self.is_cuda and self.dtype != torch.float32
+        self._mark("fwd")
         with torch.autocast("cuda", dtype=self.dtype, enabled=autocast_on):
             logits, labels = self.model(im_q, im_k)
+        self._mark("loss")
         loss = moco_loss(logits)
         self.optimizer.zero_grad(set_to_none=True)
+        self._mark("bwd")
         loss.backward()
+        self._mark("opt")
         self.optimizer.step()
+        self._mark("end")
         return loss, logits, labels
 
     def step_from_raw(self, raw: torch.Tensor):
+        self._phase_events = [] if self._timing else None
+        self._mark("aug")
         im_q, im_k = self.augment(raw)
         return self.step(im_q, im_k)
+
+    # -- optional per-phase GPU timing (HIP events; SURVEY §5.1) ------------
+    _timing = False
+    _phase_events = None
+
+    def enable_timing(self, on: bool = True):
+        self._timing = on and self.is_cuda
+
+    def _mark(self, name: str):
+        if self._timing and self._phase_events is not None:
+            ev = torch.cuda.Event(enable_timing=True)
+            ev.record()
+            self._phase_events.append((name, ev))
+
+    def phase_times_ms(self):
+        """Call after a synchronize: ms per phase of the last step."""
+        if not self._phase_events:
+            return {}
+        out = {}
+        evs = self._phase_events
+        for (name, e0), (_, e1) in zip(evs[:-1], evs[1:]):
+            out[name] = e0.elapsed_time(e1)
+        return out
 
 
 def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args):
