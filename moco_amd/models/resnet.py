@@ -18,6 +18,8 @@ from typing import Callable, List, Optional, Type, Union
 import torch
 import torch.nn as nn
 
+from ..ops.bn import bn_act
+
 
 def conv3x3(cin: int, cout: int, stride: int = 1, dilation: int = 1) -> nn.Conv2d:
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=dilation,
@@ -45,12 +47,12 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
+        out = bn_act(self.conv1(x), self.bn1, relu=True)
         if self.downsample is not None:
-            identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+            identity = bn_act(self.downsample[0](x), self.downsample[1],
+                              relu=False)
+        return bn_act(self.conv2(out), self.bn2, relu=True,
+                      residual=identity)
 
 
 class Bottleneck(nn.Module):
@@ -72,13 +74,13 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        out = bn_act(self.conv1(x), self.bn1, relu=True)
+        out = bn_act(self.conv2(out), self.bn2, relu=True)
         if self.downsample is not None:
-            identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+            identity = bn_act(self.downsample[0](x), self.downsample[1],
+                              relu=False)
+        return bn_act(self.conv3(out), self.bn3, relu=True,
+                      residual=identity)
 
 
 class ResNet(nn.Module):
@@ -132,7 +134,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.relu(self.bn1(self.conv1(x)))
+        x = bn_act(self.conv1(x), self.bn1, relu=True)
         x = self.maxpool(x)
         x = self.layer1(x)
         x = self.layer2(x)

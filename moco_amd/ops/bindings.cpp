@@ -179,6 +179,90 @@ void ema_flat(torch::Tensor dst, torch::Tensor src, double m) {
 }
 
 // ---------------------------------------------------------------------------
+// Fused BatchNorm (+ReLU, +residual).  x is a channels_last [N,C,H,W] bf16
+// tensor viewed here as [R, C] with R = N*H*W.
+
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x,
+                                        c10::optional<torch::Tensor> res,
+                                        torch::Tensor gamma,
+                                        torch::Tensor beta,
+                                        c10::optional<torch::Tensor> rmean,
+                                        c10::optional<torch::Tensor> rvar,
+                                        double eps, double momentum,
+                                        bool relu) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x));
+  const int64_t C = x.size(1);
+  const int64_t R = x.numel() / C;
+  at::cuda::CUDAGuard guard(x.device());
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto part_s = torch::empty({304, C}, f32);
+  auto part_q = torch::empty({304, C}, f32);
+  auto mean = torch::empty({C}, f32);
+  auto rstd = torch::empty({C}, f32);
+  CHECK_LAUNCH(launch_bn_stats(
+      cur_stream(), x.data_ptr(), part_s.data_ptr<float>(),
+      part_q.data_ptr<float>(), mean.data_ptr<float>(),
+      rstd.data_ptr<float>(),
+      rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
+      rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, (float)eps,
+      (float)momentum));
+  auto y = torch::empty_like(x);
+  CHECK_LAUNCH(launch_bn_apply(
+      cur_stream(), x.data_ptr(),
+      res.has_value() ? res->data_ptr() : nullptr, y.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      gamma.data_ptr<float>(), beta.data_ptr<float>(), R, C, relu));
+  return {y, mean, rstd};
+}
+
+torch::Tensor bn_fwd_eval(torch::Tensor x, c10::optional<torch::Tensor> res,
+                          torch::Tensor gamma, torch::Tensor beta,
+                          torch::Tensor mean, torch::Tensor rstd, bool relu) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x));
+  const int64_t C = x.size(1);
+  const int64_t R = x.numel() / C;
+  at::cuda::CUDAGuard guard(x.device());
+  auto y = torch::empty_like(x);
+  CHECK_LAUNCH(launch_bn_apply(
+      cur_stream(), x.data_ptr(),
+      res.has_value() ? res->data_ptr() : nullptr, y.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      gamma.data_ptr<float>(), beta.data_ptr<float>(), R, C, relu));
+  return y;
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor g, torch::Tensor y,
+                                  torch::Tensor x, torch::Tensor mean,
+                                  torch::Tensor rstd, torch::Tensor gamma,
+                                  bool relu, bool train, bool need_dres) {
+  TORCH_CHECK(g.is_cuda() && is_bf16(g));
+  const int64_t C = x.size(1);
+  const int64_t R = x.numel() / C;
+  at::cuda::CUDAGuard guard(x.device());
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto part_db = torch::empty({304, C}, f32);
+  auto part_dg = torch::empty({304, C}, f32);
+  auto dbeta = torch::empty({C}, f32);
+  auto dgamma = torch::empty({C}, f32);
+  CHECK_LAUNCH(launch_bn_bwd_reduce(
+      cur_stream(), g.data_ptr(), y.data_ptr(), x.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      part_db.data_ptr<float>(), part_dg.data_ptr<float>(),
+      dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), R, C, relu));
+  auto dx = torch::empty_like(g);
+  torch::Tensor dres;
+  if (need_dres) dres = torch::empty_like(g);
+  CHECK_LAUNCH(launch_bn_bwd_dx(
+      cur_stream(), g.data_ptr(), y.data_ptr(), x.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      gamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+      dgamma.data_ptr<float>(), dx.data_ptr(),
+      need_dres ? dres.data_ptr() : nullptr, R, C, relu, train));
+  if (need_dres) return {dx, dgamma, dbeta, dres};
+  return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------------
 torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
                                int64_t S, bool out_bf16, bool channels_last) {
   CHECK_CUDA_CONTIG(src);
@@ -219,4 +303,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("multi_tensor_sgd", &multi_tensor_sgd);
   mod.def("ema_flat", &ema_flat);
   mod.def("two_crop_augment", &two_crop_augment);
+  mod.def("bn_fwd_train", &bn_fwd_train);
+  mod.def("bn_fwd_eval", &bn_fwd_eval);
+  mod.def("bn_bwd", &bn_bwd);
 }

@@ -51,6 +51,26 @@ int launch_multi_tensor_sgd(void* stream, const int64_t* ptrs_p,
 int launch_ema_flat(void* stream, float* dst, const float* src, int64_t n,
                     float m);
 
+// ---- bn.hip ---------------------------------------------------------------
+// x/y/g/res: [R, C] bf16 (NHWC flattened); stats fp32; C in {64..256 dividing
+// 256, or multiples of 256 up to 2048}; part_* workspaces are [304, C] fp32.
+int launch_bn_stats(void* stream, const void* x, float* part_s, float* part_q,
+                    float* mean, float* rstd, float* running_mean,
+                    float* running_var, int64_t R, int64_t C, float eps,
+                    float momentum);
+int launch_bn_apply(void* stream, const void* x, const void* res, void* y,
+                    const float* mean, const float* rstd, const float* gamma,
+                    const float* beta, int64_t R, int64_t C, bool relu);
+int launch_bn_bwd_reduce(void* stream, const void* g, const void* y,
+                         const void* x, const float* mean, const float* rstd,
+                         float* part_db, float* part_dg, float* dbeta,
+                         float* dgamma, int64_t R, int64_t C, bool relu);
+int launch_bn_bwd_dx(void* stream, const void* g, const void* y,
+                     const void* x, const float* mean, const float* rstd,
+                     const float* gamma, const float* dbeta,
+                     const float* dgamma, void* dx, void* dres, int64_t R,
+                     int64_t C, bool relu, bool train);
+
 // ---- augment.hip ----------------------------------------------------------
 // src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,
 // NCHW or NHWC per flags); tmp1/tmp2: [M,3,S,S] fp32 workspaces;

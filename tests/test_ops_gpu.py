@@ -282,3 +282,93 @@ def test_moco_gpu_matches_cpu_reference_step():
     corr = torch.corrcoef(torch.stack([lcf, lgf]))[0, 1]
     assert corr > 0.999, corr
     assert (lcf - lgf).abs().max() < 0.5
+
+
+# ---------------------------------------------------------------------------
+# fused BatchNorm (+ReLU, +residual)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("C,HW,relu,res", [
+    (64, 28, True, False), (128, 14, True, True), (256, 7, False, False),
+    (512, 14, True, True), (2048, 7, True, False), (64, 112, True, False)])
+def test_fused_bn_train_fwd_bwd(C, HW, relu, res):
+    import torch.nn as nn
+    from moco_amd.ops.bn import bn_act
+    torch.manual_seed(10)
+    N = 8
+    x = torch.randn(N, C, HW, HW) * 2 + 0.3
+
+    # reference: fp32 CPU
+    bn_ref = nn.BatchNorm2d(C)
+    bn_ref.weight.data.uniform_(0.5, 1.5)
+    bn_ref.bias.data.uniform_(-0.3, 0.3)
+    xr = x.clone().requires_grad_(True)
+    rr = None
+    res_t = torch.randn(N, C, HW, HW) if res else None
+    if res:
+        rr = res_t.clone().requires_grad_(True)
+    out = bn_ref(xr)
+    if res:
+        out = out + rr
+    if relu:
+        out = torch.relu(out)
+    g = torch.randn_like(out)
+    out.backward(g)
+
+    # fused path: bf16 channels_last GPU
+    bn_gpu = nn.BatchNorm2d(C).cuda()
+    bn_gpu.load_state_dict(bn_ref.state_dict())
+    bn_gpu.running_mean.zero_()
+    bn_gpu.running_var.fill_(1.0)
+    bn_gpu.num_batches_tracked.zero_()
+    xg = x.cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    rg = None
+    if res:
+        rg = res_t.cuda().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    yg = bn_act(xg, bn_gpu, relu=relu, residual=rg)
+    yg.backward(g.cuda().to(torch.bfloat16)
+                .contiguous(memory_format=torch.channels_last))
+
+    assert (yg.float().cpu() - out.detach()).abs().max() < 8e-2
+    # running stats must match torch semantics
+    assert torch.allclose(bn_gpu.running_mean.cpu(),
+                          bn_ref.running_mean, atol=2e-2)
+    assert torch.allclose(bn_gpu.running_var.cpu(),
+                          bn_ref.running_var, atol=4e-2, rtol=1e-2)
+    assert bn_gpu.num_batches_tracked.item() == 1
+    # grads (bf16 tolerance; dx magnitudes are O(1) here)
+    M = x[:, 0].numel()
+    assert (bn_gpu.weight.grad.cpu() - bn_ref.weight.grad).abs().max() \
+        < 3e-2 * max(1.0, float(bn_ref.weight.grad.abs().max()))
+    assert (bn_gpu.bias.grad.cpu() - bn_ref.bias.grad).abs().max() \
+        < 3e-2 * max(1.0, float(bn_ref.bias.grad.abs().max()))
+    dx_err = (xg.grad.float().cpu() - xr.grad).abs().max()
+    assert dx_err < 0.11, dx_err
+    if res:
+        dr_err = (rg.grad.float().cpu() - rr.grad).abs().max()
+        assert dr_err < 5e-2, dr_err
+
+
+def test_fused_bn_eval_matches_torch():
+    import torch.nn as nn
+    from moco_amd.ops.bn import bn_act
+    torch.manual_seed(11)
+    C = 128
+    bn = nn.BatchNorm2d(C)
+    bn.running_mean.normal_()
+    bn.running_var.uniform_(0.5, 2.0)
+    bn.weight.data.normal_(1.0, 0.2)
+    bn.bias.data.normal_()
+    bn.eval()
+    x = torch.randn(4, C, 9, 9)
+    ref = torch.relu(bn(x))
+    bng = nn.BatchNorm2d(C).cuda()
+    bng.load_state_dict(bn.state_dict())
+    bng.eval()
+    xg = x.cuda().to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last)
+    y = bn_act(xg, bng, relu=True)
+    assert (y.float().cpu() - ref).abs().max() < 6e-2
+    assert bng.num_batches_tracked.item() == 0
