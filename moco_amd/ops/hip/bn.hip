@@ -97,12 +97,24 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
     float* __restrict__ mean, float* __restrict__ rstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     int64_t R, int C, int P, float eps, float momentum) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
+  // 8 partial-groups x 32 channels per block: the P-loop loads pipeline
+  // across groups instead of serializing on one thread
+  const int c = blockIdx.x * 32 + (threadIdx.x & 31);
+  const int pg = threadIdx.x >> 5;
   float s = 0.f, q = 0.f;
-  for (int p = 0; p < P; ++p) {
-    s += part_s[(int64_t)p * C + c];
-    q += part_q[(int64_t)p * C + c];
+  if (c < C)
+    for (int p = pg; p < P; p += 8) {
+      s += part_s[(int64_t)p * C + c];
+      q += part_q[(int64_t)p * C + c];
+    }
+  __shared__ float lss[256], lqq[256];
+  lss[threadIdx.x] = s;
+  lqq[threadIdx.x] = q;
+  __syncthreads();
+  if (pg != 0 || c >= C) return;
+  for (int g = 1; g < 8; ++g) {
+    s += lss[(threadIdx.x & 31) + g * 32];
+    q += lqq[(threadIdx.x & 31) + g * 32];
   }
   const float m = s / (float)R;
   const float var = fmaxf(q / (float)R - m * m, 0.f);
@@ -190,12 +202,22 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
 __global__ __launch_bounds__(256) void bn_bwd_final_kernel(
     const float* __restrict__ part_db, const float* __restrict__ part_dg,
     float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int P) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
+  const int c = blockIdx.x * 32 + (threadIdx.x & 31);
+  const int pg = threadIdx.x >> 5;
   float db = 0.f, dg = 0.f;
-  for (int p = 0; p < P; ++p) {
-    db += part_db[(int64_t)p * C + c];
-    dg += part_dg[(int64_t)p * C + c];
+  if (c < C)
+    for (int p = pg; p < P; p += 8) {
+      db += part_db[(int64_t)p * C + c];
+      dg += part_dg[(int64_t)p * C + c];
+    }
+  __shared__ float lss[256], lqq[256];
+  lss[threadIdx.x] = db;
+  lqq[threadIdx.x] = dg;
+  __syncthreads();
+  if (pg != 0 || c >= C) return;
+  for (int g = 1; g < 8; ++g) {
+    db += lss[(threadIdx.x & 31) + g * 32];
+    dg += lqq[(threadIdx.x & 31) + g * 32];
   }
   dbeta[c] = db;
   dgamma[c] = dg;
@@ -242,6 +264,16 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
   }
 }
 
+// small feature maps need few partial blocks (the finalize cost scales
+// with P); big ones want the chip full.  >=4 row-iterations per block.
+int bn_partial_blocks(int64_t R, int64_t C) {
+  const int64_t rpi = 256 / (C / 8);
+  int64_t p = R / (rpi * 4);
+  if (p > BN_PARTIAL) p = BN_PARTIAL;
+  if (p < 1) p = 1;
+  return (int)p;
+}
+
 int elementwise_grid(int64_t total8) {
   int64_t b = (total8 + 255) / 256;
   if (b > 2048) b = 2048;
@@ -266,12 +298,13 @@ extern "C" int launch_bn_stats(void* stream, const void* x, float* part_s,
                                float momentum) {
   if (!bn_c_ok(C)) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
-  bn_stats_partial_kernel<<<BN_PARTIAL, 256, 0, s>>>(
+  const int P = bn_partial_blocks(R, C);
+  bn_stats_partial_kernel<<<P, 256, 0, s>>>(
       (const unsigned short*)x, part_s, part_q, R, (int)C);
   HIP_CHECK_LAST();
-  bn_stats_final_kernel<<<dim3((unsigned)((C + 255) / 256)), 256, 0, s>>>(
+  bn_stats_final_kernel<<<dim3((unsigned)((C + 31) / 32)), 256, 0, s>>>(
       part_s, part_q, mean, rstd, running_mean, running_var, R, (int)C,
-      BN_PARTIAL, eps, momentum);
+      P, eps, momentum);
   HIP_CHECK_LAST();
   return 0;
 }
@@ -306,17 +339,18 @@ extern "C" int launch_bn_bwd_reduce(void* stream, const void* g, const void* y,
                                     bool relu) {
   if (!bn_c_ok(C)) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
+  const int P = bn_partial_blocks(R, C);
   if (relu)
-    bn_bwd_reduce_kernel<true><<<BN_PARTIAL, 256, 0, s>>>(
+    bn_bwd_reduce_kernel<true><<<P, 256, 0, s>>>(
         (const unsigned short*)g, (const unsigned short*)y,
         (const unsigned short*)x, mean, rstd, part_db, part_dg, R, (int)C);
   else
-    bn_bwd_reduce_kernel<false><<<BN_PARTIAL, 256, 0, s>>>(
+    bn_bwd_reduce_kernel<false><<<P, 256, 0, s>>>(
         (const unsigned short*)g, (const unsigned short*)y,
         (const unsigned short*)x, mean, rstd, part_db, part_dg, R, (int)C);
   HIP_CHECK_LAST();
-  bn_bwd_final_kernel<<<dim3((unsigned)((C + 255) / 256)), 256, 0, s>>>(
-      part_db, part_dg, dbeta, dgamma, (int)C, BN_PARTIAL);
+  bn_bwd_final_kernel<<<dim3((unsigned)((C + 31) / 32)), 256, 0, s>>>(
+      part_db, part_dg, dbeta, dgamma, (int)C, P);
   HIP_CHECK_LAST();
   return 0;
 }

@@ -292,27 +292,29 @@ def test_moco_gpu_matches_cpu_reference_step():
     (64, 28, True, False), (128, 14, True, True), (256, 7, False, False),
     (512, 14, True, True), (2048, 7, True, False), (64, 112, True, False)])
 def test_fused_bn_train_fwd_bwd(C, HW, relu, res):
+    """Oracle = fp32 autograd on the SAME bf16-quantized inputs the kernel
+    sees (bf16 input rounding shifts the ReLU mask for elements near zero,
+    which is a property of bf16 activations, not of the kernel — quantizing
+    the oracle's inputs removes that confound)."""
     import torch.nn as nn
     from moco_amd.ops.bn import bn_act
     torch.manual_seed(10)
     N = 8
-    x = torch.randn(N, C, HW, HW) * 2 + 0.3
+    x = (torch.randn(N, C, HW, HW) * 2 + 0.3).to(torch.bfloat16).float()
+    res_t = torch.randn(N, C, HW, HW).to(torch.bfloat16).float() \
+        if res else None
+    g = torch.randn(N, C, HW, HW).to(torch.bfloat16).float()
 
-    # reference: fp32 CPU
     bn_ref = nn.BatchNorm2d(C)
     bn_ref.weight.data.uniform_(0.5, 1.5)
     bn_ref.bias.data.uniform_(-0.3, 0.3)
     xr = x.clone().requires_grad_(True)
-    rr = None
-    res_t = torch.randn(N, C, HW, HW) if res else None
-    if res:
-        rr = res_t.clone().requires_grad_(True)
+    rr = res_t.clone().requires_grad_(True) if res else None
     out = bn_ref(xr)
     if res:
         out = out + rr
     if relu:
         out = torch.relu(out)
-    g = torch.randn_like(out)
     out.backward(g)
 
     # fused path: bf16 channels_last GPU
@@ -331,34 +333,29 @@ def test_fused_bn_train_fwd_bwd(C, HW, relu, res):
     yg.backward(g.cuda().to(torch.bfloat16)
                 .contiguous(memory_format=torch.channels_last))
 
-    # NOTE on tolerances: the fused path keeps activations bf16, so elements
-    # whose pre-ReLU value rounds across 0 get a different ReLU mask than the
-    # fp32 reference — a legitimate bf16 quantization effect, not a kernel
-    # bug.  We therefore bound the BULK error tightly and allow a tiny
-    # fraction of mask-boundary outliers.
-    def bulk_close(a, b, atol, frac=2e-4):
+    def bulk_close(a, b, atol, frac=5e-5, what=""):
         d = (a - b).abs()
-        assert float(d.mean()) < atol, ("mean", float(d.mean()))
-        assert float((d > 10 * atol).float().mean()) < frac, \
-            ("outliers", float((d > 10 * atol).float().mean()))
+        assert float(d.mean()) < atol, (what, "mean", float(d.mean()))
+        assert float((d > 20 * atol).float().mean()) < frac, \
+            (what, "outliers", float((d > 20 * atol).float().mean()))
 
-    bulk_close(yg.float().cpu(), out.detach(), 2e-2)
-    # running stats must match torch semantics
+    bulk_close(yg.float().cpu(), out.detach(), 1e-2, what="y")
     assert torch.allclose(bn_gpu.running_mean.cpu(),
-                          bn_ref.running_mean, atol=2e-2)
+                          bn_ref.running_mean, atol=1e-2)
     assert torch.allclose(bn_gpu.running_var.cpu(),
-                          bn_ref.running_var, atol=4e-2, rtol=1e-2)
+                          bn_ref.running_var, atol=2e-2, rtol=1e-2)
     assert bn_gpu.num_batches_tracked.item() == 1
     M = float(x[:, 0].numel())
-    scale = max(1.0, float(bn_ref.weight.grad.abs().max()))
-    assert (bn_gpu.weight.grad.cpu() - bn_ref.weight.grad).abs().max() \
-        < 5e-2 * scale + 2e-3 * M ** 0.5
-    scale = max(1.0, float(bn_ref.bias.grad.abs().max()))
-    assert (bn_gpu.bias.grad.cpu() - bn_ref.bias.grad).abs().max() \
-        < 5e-2 * scale + 2e-3 * M ** 0.5
-    bulk_close(xg.grad.float().cpu(), xr.grad, 2e-2)
+    # per-channel reductions of bf16-rounded terms: sqrt(M)-scaled slack
+    tol = 2e-2 * max(1.0, float(bn_ref.weight.grad.abs().max())) \
+        + 5e-3 * M ** 0.5
+    assert (bn_gpu.weight.grad.cpu() - bn_ref.weight.grad).abs().max() < tol
+    tol = 2e-2 * max(1.0, float(bn_ref.bias.grad.abs().max())) \
+        + 5e-3 * M ** 0.5
+    assert (bn_gpu.bias.grad.cpu() - bn_ref.bias.grad).abs().max() < tol
+    bulk_close(xg.grad.float().cpu(), xr.grad, 1e-2, what="dx")
     if res:
-        bulk_close(rg.grad.float().cpu(), rr.grad, 2e-2)
+        bulk_close(rg.grad.float().cpu(), rr.grad, 1e-2, what="dres")
 
 
 def test_fused_bn_eval_matches_torch():
