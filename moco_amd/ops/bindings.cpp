@@ -195,8 +195,8 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x,
   const int64_t R = x.numel() / C;
   at::cuda::CUDAGuard guard(x.device());
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto part_s = torch::empty({304, C}, f32);
-  auto part_q = torch::empty({304, C}, f32);
+  auto part_s = torch::empty({1024, C}, f32);
+  auto part_q = torch::empty({1024, C}, f32);
   auto mean = torch::empty({C}, f32);
   auto rstd = torch::empty({C}, f32);
   CHECK_LAUNCH(launch_bn_stats(
@@ -240,8 +240,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor g, torch::Tensor y,
   const int64_t R = x.numel() / C;
   at::cuda::CUDAGuard guard(x.device());
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto part_db = torch::empty({304, C}, f32);
-  auto part_dg = torch::empty({304, C}, f32);
+  auto part_db = torch::empty({1024, C}, f32);
+  auto part_dg = torch::empty({1024, C}, f32);
   auto dbeta = torch::empty({C}, f32);
   auto dgamma = torch::empty({C}, f32);
   CHECK_LAUNCH(launch_bn_bwd_reduce(

@@ -20,7 +20,8 @@
 
 namespace {
 
-constexpr int BN_PARTIAL = 304;  // ~1.2 blocks/CU on 256 CUs
+constexpr int BN_PARTIAL = 1024;  // 4 blocks/CU: memory-bound reduction
+                                  // needs ~16 waves/CU to hide HBM latency
 
 typedef unsigned short ushort8v __attribute__((ext_vector_type(8)));
 
@@ -98,13 +99,13 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
     float* __restrict__ mean, float* __restrict__ rstd,
     float* __restrict__ running_mean, float* __restrict__ running_var,
     int64_t R, int C, int P, float eps, float momentum) {
-  // 8 partial-groups x 32 channels per block: the P-loop loads pipeline
-  // across groups instead of serializing on one thread
-  const int c = blockIdx.x * 32 + (threadIdx.x & 31);
-  const int pg = threadIdx.x >> 5;
+  // 32 partial-groups x 8 channels per block: enough independent loads in
+  // flight to cover the P-deep strided reads even for small C
+  const int c = blockIdx.x * 8 + (threadIdx.x & 7);
+  const int pg = threadIdx.x >> 3;
   float s = 0.f, q = 0.f;
   if (c < C)
-    for (int p = pg; p < P; p += 8) {
+    for (int p = pg; p < P; p += 32) {
       s += part_s[(int64_t)p * C + c];
       q += part_q[(int64_t)p * C + c];
     }
@@ -113,9 +114,9 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
   lqq[threadIdx.x] = q;
   __syncthreads();
   if (pg != 0 || c >= C) return;
-  for (int g = 1; g < 8; ++g) {
-    s += lss[(threadIdx.x & 31) + g * 32];
-    q += lqq[(threadIdx.x & 31) + g * 32];
+  for (int g = 1; g < 32; ++g) {
+    s += lss[(threadIdx.x & 7) + g * 8];
+    q += lqq[(threadIdx.x & 7) + g * 8];
   }
   const float m = s / (float)R;
   const float var = fmaxf(q / (float)R - m * m, 0.f);
@@ -203,11 +204,11 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
 __global__ __launch_bounds__(256) void bn_bwd_final_kernel(
     const float* __restrict__ part_db, const float* __restrict__ part_dg,
     float* __restrict__ dbeta, float* __restrict__ dgamma, int C, int P) {
-  const int c = blockIdx.x * 32 + (threadIdx.x & 31);
-  const int pg = threadIdx.x >> 5;
+  const int c = blockIdx.x * 8 + (threadIdx.x & 7);
+  const int pg = threadIdx.x >> 3;
   float db = 0.f, dg = 0.f;
   if (c < C)
-    for (int p = pg; p < P; p += 8) {
+    for (int p = pg; p < P; p += 32) {
       db += part_db[(int64_t)p * C + c];
       dg += part_dg[(int64_t)p * C + c];
     }
@@ -216,9 +217,9 @@ __global__ __launch_bounds__(256) void bn_bwd_final_kernel(
   lqq[threadIdx.x] = dg;
   __syncthreads();
   if (pg != 0 || c >= C) return;
-  for (int g = 1; g < 8; ++g) {
-    db += lss[(threadIdx.x & 31) + g * 32];
-    dg += lqq[(threadIdx.x & 31) + g * 32];
+  for (int g = 1; g < 32; ++g) {
+    db += lss[(threadIdx.x & 7) + g * 8];
+    dg += lqq[(threadIdx.x & 7) + g * 8];
   }
   dbeta[c] = db;
   dgamma[c] = dg;
@@ -303,7 +304,7 @@ extern "C" int launch_bn_stats(void* stream, const void* x, float* part_s,
  hipLaunchKernelGGL(( bn_stats_partial_kernel), dim3(P), dim3(256), 0, s, 
       (const unsigned short*)x, part_s, part_q, R, (int)C);
   HIP_CHECK_LAST();
- hipLaunchKernelGGL(( bn_stats_final_kernel), dim3(dim3((unsigned)((C + 31) / 32))), dim3(256), 0, s, 
+ hipLaunchKernelGGL(( bn_stats_final_kernel), dim3(dim3((unsigned)((C + 7) / 8))), dim3(256), 0, s, 
       part_s, part_q, mean, rstd, running_mean, running_var, R, (int)C,
       P, eps, momentum);
   HIP_CHECK_LAST();
@@ -350,7 +351,7 @@ extern "C" int launch_bn_bwd_reduce(void* stream, const void* g, const void* y,
         (const unsigned short*)g, (const unsigned short*)y,
         (const unsigned short*)x, mean, rstd, part_db, part_dg, R, (int)C);
   HIP_CHECK_LAST();
- hipLaunchKernelGGL(( bn_bwd_final_kernel), dim3(dim3((unsigned)((C + 31) / 32))), dim3(256), 0, s, 
+ hipLaunchKernelGGL(( bn_bwd_final_kernel), dim3(dim3((unsigned)((C + 7) / 8))), dim3(256), 0, s, 
       part_db, part_dg, dbeta, dgamma, (int)C, P);
   HIP_CHECK_LAST();
   return 0;
