@@ -77,6 +77,14 @@ def get_parser():
     parser.add_argument("--save-dir", default=".", type=str)
     parser.add_argument("--max-steps-per-epoch", default=0, type=int,
                         help="truncate epochs (smoke/testing); 0 = full")
+    parser.add_argument("--dist-timeout", default=1800, type=int,
+                        help="collective timeout (s); a wedged rank aborts "
+                             "the job instead of hanging (SURVEY §5.3)")
+    parser.add_argument("--deterministic", action="store_true",
+                        help="deterministic kernels + no MIOpen benchmark "
+                             "(debug/tests; slower)")
+    parser.add_argument("--metrics-file", default="metrics.jsonl", type=str,
+                        help="JSONL metrics stream (rank 0; '' disables)")
     return parser
 
 
@@ -112,13 +120,21 @@ def main_worker(gpu, ngpus_per_node, args):
     args.gpu = gpu
     distributed = args.world_size > 1 or (args.multiprocessing_distributed
                                           and ngpus_per_node > 1)
+    if args.deterministic:
+        torch.use_deterministic_algorithms(True, warn_only=True)
+        torch.backends.cudnn.deterministic = True
+        torch.backends.cudnn.benchmark = False
     if distributed:
+        # fail fast on a wedged rank instead of hanging RCCL
+        os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
         if args.multiprocessing_distributed:
             args.rank = args.rank * ngpus_per_node + gpu if args.rank > 0 \
                 else gpu
+        from datetime import timedelta
         dist.init_process_group(backend=args.dist_backend,
                                 init_method=args.dist_url,
-                                world_size=args.world_size, rank=args.rank)
+                                world_size=args.world_size, rank=args.rank,
+                                timeout=timedelta(seconds=args.dist_timeout))
         if args.rank != 0:  # silence non-master ranks (ref main_moco.py:139)
             def print_pass(*a, **k):
                 pass
@@ -182,6 +198,14 @@ def main_worker(gpu, ngpus_per_node, args):
     stepper = MoCoStepper(model, optimizer, device,
                           aug_plus=args.aug_plus, out_size=args.crop_size,
                           dtype=dtype)
+    if args.deterministic:
+        torch.backends.cudnn.benchmark = False  # stepper turns it on
+
+    from moco_amd.utils import MetricsLogger
+    is_master = not distributed or dist.get_rank() == 0
+    metrics = MetricsLogger(
+        os.path.join(args.save_dir, args.metrics_file)
+        if (is_master and args.metrics_file) else None)
 
     for epoch in range(args.start_epoch, args.epochs):
         if sampler is not None:
@@ -198,9 +222,9 @@ def main_worker(gpu, ngpus_per_node, args):
 
                 def __len__(self):
                     return args.max_steps_per_epoch
-            train_epoch(_L(), stepper, epoch, args)
+            train_epoch(_L(), stepper, epoch, args, metrics=metrics)
         else:
-            train_epoch(loader, stepper, epoch, args)
+            train_epoch(loader, stepper, epoch, args, metrics=metrics)
 
         if not distributed or dist.get_rank() == 0:
             raw_model = model.module if distributed else model

@@ -124,12 +124,19 @@ class MoCo(nn.Module):
             step = self._step
             self._step += 1
 
+        with torch.no_grad():
+            # start the key-batch exchange on the comm stream FIRST so the
+            # xGMI transfer overlaps the query forward (join at key forward)
+            shuffle_job = parallel.begin_shuffle(im_k, step,
+                                                 self.shuffle_seed)
+            # EMA before the key forward, once per step (reference invariant)
+            self._momentum_update_key_encoder()
+
         q = self.encoder_q(im_q)
         q = l2_normalize(q, dim=1)
 
         with torch.no_grad():
-            self._momentum_update_key_encoder()
-            im_k_sh, idx_unshuffle = self._batch_shuffle(im_k, step)
+            im_k_sh, idx_unshuffle = shuffle_job.wait()
             k = self.encoder_k(im_k_sh)
             k = l2_normalize(k, dim=1)
             k = self._batch_unshuffle(k, idx_unshuffle)

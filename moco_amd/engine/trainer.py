@@ -102,7 +102,8 @@ class MoCoStepper:
         return out
 
 
-def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args):
+def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args,
+                metrics=None):
     """One pretraining epoch (reference train(), main_moco.py:290-335)."""
     batch_time = AverageMeter("Time", ":6.3f")
     data_time = AverageMeter("Data", ":6.3f")
@@ -114,6 +115,7 @@ def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args):
                              prefix="Epoch: [{}]".format(epoch))
 
     stepper.model.train()
+    total_images = 0
     end = time.time()
     for i, (raw, _) in enumerate(train_loader):
         data_time.update(time.time() - end)
@@ -127,7 +129,19 @@ def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args):
 
         batch_time.update(time.time() - end)
         end = time.time()
+        total_images += raw.size(0)
 
         if i % args.print_freq == 0:
             progress.display(i)
+            if metrics is not None:
+                metrics.log(epoch=epoch, step=i, loss=round(losses.val, 5),
+                            acc1=round(top1.val, 3), acc5=round(top5.val, 3),
+                            img_s=round(raw.size(0) /
+                                        max(batch_time.val, 1e-9), 2),
+                            data_s=round(data_time.val, 4))
+    if metrics is not None:
+        metrics.log(epoch=epoch, summary=True, loss_avg=round(losses.avg, 5),
+                    acc1_avg=round(top1.avg, 3),
+                    img_s_avg=round(total_images /
+                                    max(batch_time.sum, 1e-9), 2))
     return losses.avg

@@ -1,5 +1,6 @@
 from .comm import (
     all_to_all_rows,
+    begin_shuffle,
     concat_all_gather,
     get_rank,
     get_world_size,
@@ -18,4 +19,5 @@ __all__ = [
     "shuffle_exchange",
     "unshuffle_exchange",
     "all_to_all_rows",
+    "begin_shuffle",
 ]

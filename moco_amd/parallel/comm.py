@@ -153,3 +153,68 @@ def unshuffle_exchange(x: torch.Tensor,
         return x[idx_unshuffle.to(x.device)]
     dest_perm = torch.argsort(idx_unshuffle)
     return all_to_all_rows(x, dest_perm)
+
+
+# ---------------------------------------------------------------------------
+# Comm/compute overlap: run the shuffle-BN exchange (C3) on a dedicated HIP
+# stream, launched BEFORE the query-encoder forward so the xGMI transfer
+# hides under compute (SURVEY §5.8 overlap requirement).  The join point is
+# the key-encoder forward.
+# ---------------------------------------------------------------------------
+
+_COMM_STREAM = None
+
+
+def _comm_stream():
+    global _COMM_STREAM
+    if _COMM_STREAM is None:
+        _COMM_STREAM = torch.cuda.Stream()
+    return _COMM_STREAM
+
+
+class _ReadyJob:
+    def __init__(self, result):
+        self._result = result
+
+    def wait(self):
+        return self._result
+
+
+class _AsyncShuffleJob:
+    def __init__(self, out, idx_unshuffle, done_event):
+        self._out = out
+        self._idx = idx_unshuffle
+        self._done = done_event
+
+    def wait(self):
+        cur = torch.cuda.current_stream()
+        cur.wait_event(self._done)
+        # out was allocated on the comm stream; register its use here so the
+        # caching allocator doesn't recycle it under the main stream
+        self._out.record_stream(cur)
+        return self._out, self._idx
+
+
+def overlap_enabled() -> bool:
+    import os
+    return os.environ.get("MOCO_AMD_NO_OVERLAP", "0") != "1"
+
+
+@torch.no_grad()
+def begin_shuffle(x: torch.Tensor, step: int, seed: int):
+    """Start the shuffle-BN exchange; returns a job whose .wait() yields
+    (shuffled rows, idx_unshuffle)."""
+    can_overlap = (x.is_cuda and is_distributed() and get_world_size() > 1
+                   and _backend_supports_a2a() and overlap_enabled())
+    if not can_overlap:
+        return _ReadyJob(shuffle_exchange(x, step, seed))
+    stream = _comm_stream()
+    ready = torch.cuda.Event()
+    ready.record()  # x is valid once current-stream work is done
+    done = torch.cuda.Event()
+    with torch.cuda.stream(stream):
+        stream.wait_event(ready)
+        x.record_stream(stream)  # keep x alive for the comm stream
+        out, idx_unshuffle = shuffle_exchange(x, step, seed)
+        done.record()
+    return _AsyncShuffleJob(out, idx_unshuffle, done)
