@@ -78,7 +78,7 @@ def all_to_all_rows(x: torch.Tensor, dest_perm: torch.Tensor) -> torch.Tensor:
     """
     world = get_world_size()
     n = x.shape[0]
-    if world == 1:
+    if not is_distributed():
         return x[torch.argsort(dest_perm)] if dest_perm is not None else x
 
     rank = get_rank()
@@ -133,7 +133,7 @@ def shuffle_exchange(x: torch.Tensor, step: int, seed: int
     n = x.shape[0]
     idx_shuffle, idx_unshuffle = step_permutation(step, world * n, seed,
                                                   x.device)
-    if world == 1:
+    if not is_distributed():
         return x[idx_shuffle.to(x.device)], idx_unshuffle
     # global source row g must land at global slot p where
     # idx_shuffle[p] == g  =>  dest_perm = argsort? idx_shuffle maps slot->src,
@@ -149,7 +149,7 @@ def unshuffle_exchange(x: torch.Tensor,
     builder.py:106-123)."""
     world = get_world_size()
     n = x.shape[0]
-    if world == 1:
+    if not is_distributed():
         return x[idx_unshuffle.to(x.device)]
     dest_perm = torch.argsort(idx_unshuffle)
     return all_to_all_rows(x, dest_perm)
@@ -204,7 +204,7 @@ def overlap_enabled() -> bool:
 def begin_shuffle(x: torch.Tensor, step: int, seed: int):
     """Start the shuffle-BN exchange; returns a job whose .wait() yields
     (shuffled rows, idx_unshuffle)."""
-    can_overlap = (x.is_cuda and is_distributed() and get_world_size() > 1
+    can_overlap = (x.is_cuda and is_distributed()
                    and _backend_supports_a2a() and overlap_enabled())
     if not can_overlap:
         return _ReadyJob(shuffle_exchange(x, step, seed))
