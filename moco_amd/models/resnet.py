@@ -19,6 +19,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.bn import bn_act
+from ..ops.pool import stem_maxpool
 
 
 def conv3x3(cin: int, cout: int, stride: int = 1, dilation: int = 1) -> nn.Conv2d:
@@ -135,7 +136,7 @@ class ResNet(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = bn_act(self.conv1(x), self.bn1, relu=True)
-        x = self.maxpool(x)
+        x = stem_maxpool(x)
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)

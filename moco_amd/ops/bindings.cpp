@@ -263,6 +263,33 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor g, torch::Tensor y,
 }
 
 // ---------------------------------------------------------------------------
+// MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x) && x.dim() == 4);
+  const int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int64_t Ho = (H + 1) / 2, Wo = (W + 1) / 2;
+  at::cuda::CUDAGuard guard(x.device());
+  auto y = torch::empty({N, C, Ho, Wo}, x.options())
+               .contiguous(torch::MemoryFormat::ChannelsLast);
+  auto idx = torch::empty({N, Ho, Wo, C}, x.options().dtype(torch::kUInt8));
+  CHECK_LAUNCH(launch_maxpool_fwd(cur_stream(), x.data_ptr(), y.data_ptr(),
+                                  idx.data_ptr(), N, C, H, W, Ho, Wo));
+  return {y, idx};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor g, torch::Tensor idx, int64_t H,
+                          int64_t W) {
+  TORCH_CHECK(g.is_cuda() && is_bf16(g) && g.dim() == 4);
+  const int64_t N = g.size(0), C = g.size(1), Ho = g.size(2), Wo = g.size(3);
+  at::cuda::CUDAGuard guard(g.device());
+  auto dx = torch::empty({N, C, H, W}, g.options())
+                .contiguous(torch::MemoryFormat::ChannelsLast);
+  CHECK_LAUNCH(launch_maxpool_bwd(cur_stream(), g.data_ptr(), idx.data_ptr(),
+                                  dx.data_ptr(), N, C, H, W, Ho, Wo));
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
 torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
                                int64_t S, bool out_bf16, bool channels_last) {
   CHECK_CUDA_CONTIG(src);
@@ -306,4 +333,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_fwd_train", &bn_fwd_train);
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
+  mod.def("maxpool_fwd", &maxpool_fwd);
+  mod.def("maxpool_bwd", &maxpool_bwd);
 }

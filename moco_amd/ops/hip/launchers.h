@@ -82,3 +82,18 @@ int launch_two_crop_augment(void* stream, const float* src,
                             bool channels_last);
 
 }  // extern "C"
+
+// ---- pool.hip -------------------------------------------------------------
+#ifdef __cplusplus
+extern "C" {
+#endif
+// 3x3 stride-2 pad-1 max pool, NHWC bf16; idx: per-output uint8 tap id [0,9)
+int launch_maxpool_fwd(void* stream, const void* x, void* y, void* idx,
+                       int64_t N, int64_t C, int64_t H, int64_t W, int64_t Ho,
+                       int64_t Wo);
+int launch_maxpool_bwd(void* stream, const void* g, const void* idx, void* dx,
+                       int64_t N, int64_t C, int64_t H, int64_t W, int64_t Ho,
+                       int64_t Wo);
+#ifdef __cplusplus
+}
+#endif

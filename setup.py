@@ -24,6 +24,7 @@ sources = [
     "moco_amd/ops/hip/multi_tensor.hip",
     "moco_amd/ops/hip/queue.hip",
     "moco_amd/ops/hip/bn.hip",
+    "moco_amd/ops/hip/pool.hip",
     "moco_amd/ops/hip/augment.hip",
 ]
 

@@ -379,3 +379,31 @@ def test_fused_bn_eval_matches_torch():
     y = bn_act(xg, bng, relu=True)
     assert (y.float().cpu() - ref).abs().max() < 6e-2
     assert bng.num_batches_tracked.item() == 0
+
+
+# ---------------------------------------------------------------------------
+# stem max-pool
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("N,C,H", [(4, 64, 112), (2, 64, 56), (3, 128, 27)])
+def test_maxpool_fwd_bwd(N, C, H):
+    from moco_amd.ops.pool import stem_maxpool
+    torch.manual_seed(12)
+    x = torch.randn(N, C, H, H).to(torch.bfloat16).float()
+    xr = x.clone().requires_grad_(True)
+    ref = torch.nn.functional.max_pool2d(xr, 3, stride=2, padding=1)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+
+    xg = x.cuda().to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y = stem_maxpool(xg)
+    assert y.shape == ref.shape
+    y.backward(g.cuda().to(torch.bfloat16)
+               .contiguous(memory_format=torch.channels_last))
+    assert (y.float().cpu() - ref.detach()).abs().max() < 1e-2
+    # ties can route grad to a different (equal) tap; bf16 rounding can also
+    # merge near-equal values: bound the bulk error
+    d = (xg.grad.float().cpu() - xr.grad).abs()
+    assert float(d.mean()) < 2e-3, float(d.mean())
+    assert float((d > 0.2).float().mean()) < 1e-3
