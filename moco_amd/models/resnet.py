@@ -19,6 +19,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.bn import bn_act
+from ..ops.conv1x1 import conv1x1 as conv1x1_op
 from ..ops.pool import stem_maxpool
 
 
@@ -75,12 +76,12 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
-        out = bn_act(self.conv1(x), self.bn1, relu=True)
+        out = bn_act(conv1x1_op(x, self.conv1), self.bn1, relu=True)
         out = bn_act(self.conv2(out), self.bn2, relu=True)
         if self.downsample is not None:
             identity = bn_act(self.downsample[0](x), self.downsample[1],
                               relu=False)
-        return bn_act(self.conv3(out), self.bn3, relu=True,
+        return bn_act(conv1x1_op(out, self.conv3), self.bn3, relu=True,
                       residual=identity)
 
 

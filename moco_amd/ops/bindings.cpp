@@ -263,6 +263,20 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor g, torch::Tensor y,
 }
 
 // ---------------------------------------------------------------------------
+// bf16 NT-GEMM (1x1-conv forward/dgrad): C[M,N] = A[M,K] @ Bt[N,K]^T
+torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt) {
+  CHECK_CUDA_CONTIG(A);
+  CHECK_CUDA_CONTIG(Bt);
+  TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && A.size(1) == Bt.size(1));
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  at::cuda::CUDAGuard guard(A.device());
+  auto C = torch::empty({M, N}, A.options());
+  CHECK_LAUNCH(launch_gemm_nt_bf16(cur_stream(), A.data_ptr(), Bt.data_ptr(),
+                                   C.data_ptr(), M, N, K));
+  return C;
+}
+
+// ---------------------------------------------------------------------------
 // MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && is_bf16(x) && x.dim() == 4);
@@ -334,5 +348,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
   mod.def("maxpool_fwd", &maxpool_fwd);
+  mod.def("gemm_nt_bf16", &gemm_nt_bf16);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

@@ -97,3 +97,11 @@ int launch_maxpool_bwd(void* stream, const void* g, const void* idx, void* dx,
 #ifdef __cplusplus
 }
 #endif
+
+// ---- conv1x1.hip ----------------------------------------------------------
+#ifdef __cplusplus
+extern "C"
+#endif
+// C[M,N] = A[M,K] @ Bt[N,K]^T, bf16 in/out, fp32 accumulate (1x1-conv GEMM)
+int launch_gemm_nt_bf16(void* stream, const void* A, const void* Bt, void* C,
+                        int64_t M, int64_t N, int64_t K);

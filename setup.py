@@ -25,6 +25,7 @@ sources = [
     "moco_amd/ops/hip/queue.hip",
     "moco_amd/ops/hip/bn.hip",
     "moco_amd/ops/hip/pool.hip",
+    "moco_amd/ops/hip/conv1x1.hip",
     "moco_amd/ops/hip/augment.hip",
 ]
 
