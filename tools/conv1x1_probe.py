@@ -1,6 +1,7 @@
 """A/B the custom MFMA 1x1-conv GEMM vs MIOpen/CK per ResNet-50 shape.
 Run on the GPU box: python tools/conv1x1_probe.py"""
-import time
+import os, sys, time
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
