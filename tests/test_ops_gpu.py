@@ -407,3 +407,46 @@ def test_maxpool_fwd_bwd(N, C, H):
     d = (xg.grad.float().cpu() - xr.grad).abs()
     assert float(d.mean()) < 2e-3, float(d.mean())
     assert float((d > 0.2).float().mean()) < 1e-3
+
+
+# ---------------------------------------------------------------------------
+# 1x1-conv NT-GEMM (flag-gated path; numerics kept green for round 2)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [(3136, 256, 64), (784, 64, 256),
+                                   (100, 16, 32)])
+def test_gemm_nt_bf16(M, N, K):
+    from moco_amd import _C
+    torch.manual_seed(13)
+    A = (torch.randn(M, K) * 0.5).cuda().to(torch.bfloat16)
+    Bt = (torch.randn(N, K) * 0.1).cuda().to(torch.bfloat16)
+    C = _C.gemm_nt_bf16(A, Bt)
+    ref = A.float() @ Bt.float().t()
+    assert (C.float() - ref).abs().max() < 0.05 * max(1.0, K ** 0.5 * 0.05)
+
+
+def test_conv1x1_function_matches_conv2d():
+    import os
+    os.environ["MOCO_AMD_CUSTOM_CONV1X1"] = "1"
+    try:
+        import torch.nn as nn
+        from moco_amd.ops.conv1x1 import conv1x1
+        torch.manual_seed(14)
+        conv = nn.Conv2d(64, 128, 1, bias=False).cuda() \
+            .to(memory_format=torch.channels_last)
+        x = torch.randn(2, 64, 14, 14).cuda().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+        y = conv1x1(x, conv)
+        ref = torch.nn.functional.conv2d(x.detach().float(),
+                                         conv.weight.float())
+        assert (y.float() - ref).abs().max() < 5e-2
+        g = torch.randn_like(y)
+        y.backward(g)
+        xr = x.detach().float().requires_grad_(True)
+        wr = conv.weight.detach().float().requires_grad_(True)
+        torch.nn.functional.conv2d(xr, wr).backward(g.float())
+        assert (x.grad.float() - xr.grad).abs().max() < 5e-2
+        assert (conv.weight.grad.float() - wr.grad).abs().max() \
+            < 0.1 * max(1.0, float(wr.grad.abs().max()))
+    finally:
+        os.environ.pop("MOCO_AMD_CUSTOM_CONV1X1", None)
