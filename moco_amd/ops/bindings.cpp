@@ -271,8 +271,10 @@ torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt) {
   const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
-  CHECK_LAUNCH(launch_gemm_nt_bf16(cur_stream(), A.data_ptr(), Bt.data_ptr(),
-                                   C.data_ptr(), M, N, K));
+  // LDS-staged kernel for conforming shapes, naive fragment-load fallback
+  CHECK_LAUNCH(launch_gemm_nt_lds_bf16(cur_stream(), A.data_ptr(),
+                                       Bt.data_ptr(), C.data_ptr(), M, N,
+                                       K));
   return C;
 }
 

@@ -72,3 +72,126 @@ extern "C" int launch_gemm_nt_bf16(void* stream, const void* A,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// LDS-staged variant (guide §5 ladder step-3 structure): 128x128 tile,
+// BK=64, double-buffered LDS filled by global_load_lds dwordx4, XOR-swizzled
+// images (T2, rule 21: linear LDS dest + pre-swizzled SOURCE + swizzled
+// read), 4 waves in a 2x2 wave grid, 32 v_mfma_f32_16x16x32_bf16 per K-tile.
+// Requires M % 128 == 0, N % 128 == 0, K % 64 == 0 (all ResNet-50 1x1/s1
+// shapes at batch 256 qualify); the launcher falls back to the naive kernel
+// otherwise.
+// ---------------------------------------------------------------------------
+namespace {
+
+constexpr int BM = 128, BN = 128, BK = 64;
+// one tile image: [128 rows][64 k] bf16 = 16 KiB = 1024 16-byte pieces
+
+__device__ __forceinline__ unsigned swz(unsigned row, unsigned kbyte) {
+  return kbyte ^ ((row & 7) << 4);
+}
+
+// stage one 16KB tile (A or B) into LDS via glds; `src` is the tile's
+// global base (row-major, ldK elements per row), smem the LDS byte base.
+__device__ __forceinline__ void stage_tile(const bf16* __restrict__ src,
+                                           int64_t ldK, char* smem,
+                                           int wave, int lane) {
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int piece = c * 256 + wave * 64 + lane;
+    const int row = piece / 8;
+    const int kb = (piece % 8) * 16;          // byte offset within the row
+    const int src_elem = swz(row, kb) / 2;    // pre-swizzled source column
+    // per-wave uniform LDS base; the DMA adds lane*16 itself
+    char* lds_base = smem + (int64_t)(c * 256 + wave * 64) * 16;
+    // C-style casts lower to addrspacecast (reinterpret_cast is rejected)
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)(
+            src + (int64_t)row * ldK + src_elem),
+        (__attribute__((address_space(3))) void*)lds_base, 16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ bf16x8 lds_frag(const char* smem, int row,
+                                           int kbyte) {
+  return *reinterpret_cast<const bf16x8*>(smem + row * (BK * 2) +
+                                          swz(row, kbyte));
+}
+
+__global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    bf16* __restrict__ C, int64_t M, int N, int K) {
+  __shared__ __attribute__((aligned(16))) char smem[4 * 16 * 1024];
+  // buffers: [buf][A|B] at smem + (buf*2 + ab) * 16KB
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid, 64x64 each
+  const int64_t m0 = (int64_t)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  f32x4 acc[4][4] = {};
+
+  stage_tile(A + m0 * K, K, smem, wave, lane);
+  stage_tile(Bt + (int64_t)n0 * K, K, smem + 16 * 1024, wave, lane);
+  __syncthreads();  // hipcc emits the vmcnt(0) drain for the in-flight DMA
+
+  int cur = 0;
+  const int ntiles = K / BK;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      char* nbuf = smem + ((cur ^ 1) * 2) * 16 * 1024;
+      stage_tile(A + m0 * K + (t + 1) * BK, K, nbuf, wave, lane);
+      stage_tile(Bt + (int64_t)n0 * K + (t + 1) * BK, K, nbuf + 16 * 1024,
+                 wave, lane);
+    }
+    const char* abuf = smem + (cur * 2) * 16 * 1024;
+    const char* bbuf = abuf + 16 * 1024;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[4], bf[4];
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr)
+        af[mr] = lds_frag(abuf, wr * 64 + mr * 16 + (lane & 15), kbyte);
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr)
+        bf[nr] = lds_frag(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr)
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mr], bf[nr], acc[mr][nr], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      const int col = n0 + wc * 64 + nr * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int64_t row = m0 + wr * 64 + mr * 16 + (lane >> 4) * 4 + r;
+        C[row * N + col] = f2bf(acc[mr][nr][r]);
+      }
+    }
+}
+
+}  // namespace
+
+extern "C" int launch_gemm_nt_lds_bf16(void* stream, const void* A,
+                                       const void* Bt, void* C, int64_t M,
+                                       int64_t N, int64_t K) {
+  if (M % 128 != 0 || N % 128 != 0 || K % 64 != 0)
+    return launch_gemm_nt_bf16(stream, A, Bt, C, M, N, K);
+  dim3 grid((unsigned)(M / 128), (unsigned)(N / 128));
+  gemm_nt_lds_kernel<<<grid, 256, 0, (hipStream_t)stream>>>(
+      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K);
+  HIP_CHECK_LAST();
+  return 0;
+}

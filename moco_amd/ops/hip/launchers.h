@@ -105,3 +105,8 @@ extern "C"
 // C[M,N] = A[M,K] @ Bt[N,K]^T, bf16 in/out, fp32 accumulate (1x1-conv GEMM)
 int launch_gemm_nt_bf16(void* stream, const void* A, const void* Bt, void* C,
                         int64_t M, int64_t N, int64_t K);
+#ifdef __cplusplus
+extern "C"
+#endif
+int launch_gemm_nt_lds_bf16(void* stream, const void* A, const void* Bt,
+                            void* C, int64_t M, int64_t N, int64_t K);
