@@ -84,20 +84,20 @@ extern "C" int launch_gemm_nt_bf16(void* stream, const void* A,
 // ---------------------------------------------------------------------------
 namespace {
 
-constexpr int BM = 128, BN = 128, BK = 64;
-// one tile image: [128 rows][64 k] bf16 = 16 KiB = 1024 16-byte pieces
+constexpr int BM = 128, BK = 64;
+// tile images: A [128 rows][64 k] bf16 = 16 KiB; B [BNT rows][64 k]
 
 __device__ __forceinline__ unsigned swz(unsigned row, unsigned kbyte) {
   return kbyte ^ ((row & 7) << 4);
 }
 
-// stage one 16KB tile (A or B) into LDS via glds; `src` is the tile's
-// global base (row-major, ldK elements per row), smem the LDS byte base.
+// stage a [ROWS][64] bf16 tile (ROWS*128 bytes) into LDS via glds.
+template <int ROWS>
 __device__ __forceinline__ void stage_tile(const bf16* __restrict__ src,
                                            int64_t ldK, char* smem,
                                            int wave, int lane) {
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < ROWS / 32; ++c) {
     const int piece = c * 256 + wave * 64 + lane;
     const int row = piece / 8;
     const int kb = (piece % 8) * 16;          // byte offset within the row
@@ -118,65 +118,73 @@ __device__ __forceinline__ bf16x8 lds_frag(const char* smem, int row,
                                           swz(row, kbyte));
 }
 
+// BNT = 128: 2x2 wave grid, 64x64 per wave (MR=NR=4)
+// BNT =  64: 4x1 wave grid, 32x64 per wave (MR=2, NR=4)
+// DBUF: double-buffered staging (off for single-K-tile problems)
+template <int BNT, bool DBUF>
 __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     bf16* __restrict__ C, int64_t M, int N, int K) {
-  __shared__ __attribute__((aligned(16))) char smem[4 * 16 * 1024];
-  // buffers: [buf][A|B] at smem + (buf*2 + ab) * 16KB
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int MR = BNT == 128 ? 4 : 2;
+  constexpr int ABYTES = BM * BK * 2;    // 16 KiB
+  constexpr int BBYTES = BNT * BK * 2;   // 16 or 8 KiB
   const int lane = threadIdx.x % WAVE;
   const int wave = threadIdx.x / WAVE;
-  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid, 64x64 each
+  const int wr = BNT == 128 ? (wave >> 1) : wave;
+  const int wc = BNT == 128 ? (wave & 1) : 0;
+  const int wrows = BNT == 128 ? 64 : 32;
   const int64_t m0 = (int64_t)blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  const int n0 = blockIdx.y * BNT;
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[MR][4] = {};
 
-  stage_tile(A + m0 * K, K, smem, wave, lane);
-  stage_tile(Bt + (int64_t)n0 * K, K, smem + 16 * 1024, wave, lane);
+  stage_tile<BM>(A + m0 * K, K, smem, wave, lane);
+  stage_tile<BNT>(Bt + (int64_t)n0 * K, K, smem + ABYTES, wave, lane);
   __syncthreads();  // hipcc emits the vmcnt(0) drain for the in-flight DMA
 
   int cur = 0;
   const int ntiles = K / BK;
   for (int t = 0; t < ntiles; ++t) {
-    if (t + 1 < ntiles) {
-      char* nbuf = smem + ((cur ^ 1) * 2) * 16 * 1024;
-      stage_tile(A + m0 * K + (t + 1) * BK, K, nbuf, wave, lane);
-      stage_tile(Bt + (int64_t)n0 * K + (t + 1) * BK, K, nbuf + 16 * 1024,
-                 wave, lane);
+    if (DBUF && t + 1 < ntiles) {
+      char* nbuf = smem + (cur ^ 1) * (ABYTES + BBYTES);
+      stage_tile<BM>(A + m0 * K + (t + 1) * BK, K, nbuf, wave, lane);
+      stage_tile<BNT>(Bt + (int64_t)n0 * K + (t + 1) * BK, K, nbuf + ABYTES,
+                      wave, lane);
     }
-    const char* abuf = smem + (cur * 2) * 16 * 1024;
-    const char* bbuf = abuf + 16 * 1024;
+    const char* abuf = smem + cur * (ABYTES + BBYTES);
+    const char* bbuf = abuf + ABYTES;
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      bf16x8 af[4], bf[4];
+      bf16x8 af[MR], bf[4];
       const int kbyte = ks * 64 + (lane >> 4) * 16;
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr)
-        af[mr] = lds_frag(abuf, wr * 64 + mr * 16 + (lane & 15), kbyte);
+      for (int mr = 0; mr < MR; ++mr)
+        af[mr] = lds_frag(abuf, wr * wrows + mr * 16 + (lane & 15), kbyte);
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr)
         bf[nr] = lds_frag(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int mr = 0; mr < 4; ++mr)
+      for (int mr = 0; mr < MR; ++mr)
 #pragma unroll
         for (int nr = 0; nr < 4; ++nr)
           acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mr], bf[nr], acc[mr][nr], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();
-    cur ^= 1;
+    if (DBUF || t + 1 < ntiles) __syncthreads();
+    if (DBUF) cur ^= 1;
   }
 
 #pragma unroll
-  for (int mr = 0; mr < 4; ++mr)
+  for (int mr = 0; mr < MR; ++mr)
 #pragma unroll
     for (int nr = 0; nr < 4; ++nr) {
       const int col = n0 + wc * 64 + nr * 16 + (lane & 15);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int64_t row = m0 + wr * 64 + mr * 16 + (lane >> 4) * 4 + r;
+        const int64_t row = m0 + wr * wrows + mr * 16 + (lane >> 4) * 4 + r;
         C[row * N + col] = f2bf(acc[mr][nr][r]);
       }
     }
@@ -187,11 +195,23 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
 extern "C" int launch_gemm_nt_lds_bf16(void* stream, const void* A,
                                        const void* Bt, void* C, int64_t M,
                                        int64_t N, int64_t K) {
-  if (M % 128 != 0 || N % 128 != 0 || K % 64 != 0)
+  const bool n128 = (N % 128 == 0);
+  const bool n64 = (N % 64 == 0);
+  if (M % 128 != 0 || K % 64 != 0 || !(n128 || n64))
     return launch_gemm_nt_bf16(stream, A, Bt, C, M, N, K);
-  dim3 grid((unsigned)(M / 128), (unsigned)(N / 128));
-  gemm_nt_lds_kernel<<<grid, 256, 0, (hipStream_t)stream>>>(
-      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K);
+  hipStream_t s = (hipStream_t)stream;
+  const int BNT = n128 ? 128 : 64;
+  const bool dbuf = (K / 64) > 1;
+  const int smem = (16384 + BNT * 128) * (dbuf ? 2 : 1);
+  dim3 grid((unsigned)(M / 128), (unsigned)(N / BNT));
+#define GO(BNTv, DB)                                                       \
+  gemm_nt_lds_kernel<BNTv, DB><<<grid, 256, smem, s>>>(                    \
+      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K)
+  if (BNT == 128 && dbuf) GO(128, true);
+  else if (BNT == 128) GO(128, false);
+  else if (dbuf) GO(64, true);
+  else GO(64, false);
+#undef GO
   HIP_CHECK_LAST();
   return 0;
 }
