@@ -25,7 +25,8 @@ from torch.utils.data.distributed import DistributedSampler
 
 from moco_amd.loader import build_dataset, cpu_two_crop_eval_transform
 from moco_amd.models.resnet import MODEL_REGISTRY, build_model
-from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
+from moco_amd.ops.augment import (sample_two_crop_params,
+                                  single_crop_augment)
 from moco_amd.utils import (AverageMeter, ProgressMeter, accuracy,
                             adjust_learning_rate, load_pretrained_backbone,
                             sanity_check, save_checkpoint)
@@ -59,6 +60,8 @@ def get_parser():
     p.add_argument("--pretrained", default="", type=str,
                    help="path to moco pretrained checkpoint")
     p.add_argument("--device", default=None, choices=["cuda", "cpu"])
+    p.add_argument("--lincls-dtype", default="fp32", choices=["fp32", "bf16"],
+                   help="bf16 autocast for the frozen-backbone forward")
     p.add_argument("--crop-size", default=224, type=int)
     p.add_argument("--load-size", default=256, type=int)
     p.add_argument("--synthetic-len", default=1024, type=int)
@@ -68,9 +71,11 @@ def get_parser():
     return p
 
 
-def train_transform_batch(raw: torch.Tensor, crop: int, seed_step=[0]):
+def train_transform_batch(raw: torch.Tensor, crop: int,
+                          dtype=None, channels_last=False):
     """RandomResizedCrop + flip + normalize on a raw batch (no jitter/blur —
-    reference lincls train aug)."""
+    reference lincls train aug).  Runs on whatever device `raw` lives on
+    (the HIP augmentation kernels when on GPU)."""
     B, _, H, W = raw.shape
     params = sample_two_crop_params(B, H, W, crop, aug_plus=False)
     params = params[:B]
@@ -79,7 +84,8 @@ def train_transform_batch(raw: torch.Tensor, crop: int, seed_step=[0]):
     params[:, 7:10] = 1.0  # brightness/contrast/saturation factors
     params[:, 10] = 0.0  # hue
     params[:, 6] = 0.0   # sigma
-    return _augment_torch(raw, params, crop)
+    return single_crop_augment(raw, crop, params.to(raw.device),
+                               out_dtype=dtype, channels_last=channels_last)
 
 
 def main(argv=None):
@@ -123,6 +129,8 @@ def main(argv=None):
         print(f"=> loaded pre-trained model '{args.pretrained}'")
 
     model = model.to(device)
+    if args.device == "cuda":
+        model = model.to(memory_format=torch.channels_last)
     if distributed:
         kw = dict(broadcast_buffers=False)
         if args.device == "cuda":
@@ -194,13 +202,18 @@ def train(loader, model, criterion, optimizer, epoch, device, args):
                              prefix=f"Epoch: [{epoch}]")
     # eval mode on purpose: frozen BN uses running stats (ref ≈:300)
     model.eval()
+    use_bf16 = args.lincls_dtype == "bf16" and device.type == "cuda"
     for i, (raw, target) in enumerate(loader):
         if args.max_steps_per_epoch and i >= args.max_steps_per_epoch:
             break
-        images = train_transform_batch(raw, args.crop_size).to(device)
+        images = train_transform_batch(
+            raw.to(device, non_blocking=True), args.crop_size,
+            dtype=torch.bfloat16 if use_bf16 else None,
+            channels_last=(device.type == "cuda"))
         target = target.to(device)
-        output = model(images)
-        loss = criterion(output, target)
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
+            output = model(images)
+        loss = criterion(output.float(), target)
         acc1, acc5 = accuracy(output, target, topk=(1, 5))
         losses.update(loss.item(), images.size(0))
         top1.update(acc1[0].item(), images.size(0))

@@ -46,6 +46,8 @@ class MoCoStepper:
         self.dtype = dtype if self.is_cuda else torch.float32
         self.channels_last = channels_last and self.is_cuda
         self.aug_generator = aug_generator
+        self._timing = False
+        self._phase_events = None
 
     def augment(self, raw: torch.Tensor):
         raw = raw.to(self.device, non_blocking=True)
@@ -79,9 +81,6 @@ class MoCoStepper:
         return self.step(im_q, im_k)
 
     # -- optional per-phase GPU timing (HIP events) ------------
-    _timing = False
-    _phase_events = None
-
     def enable_timing(self, on: bool = True):
         self._timing = on and self.is_cuda
 

@@ -277,3 +277,24 @@ def _augment_torch(images: torch.Tensor, params: torch.Tensor,
     mean = torch.tensor(IMAGENET_MEAN, device=x.device).view(1, 3, 1, 1)
     std = torch.tensor(IMAGENET_STD, device=x.device).view(1, 3, 1, 1)
     return (x - mean) / std
+
+
+def single_crop_augment(images: torch.Tensor, out_size: int,
+                        params: torch.Tensor,
+                        out_dtype: Optional[torch.dtype] = None,
+                        channels_last: bool = False) -> torch.Tensor:
+    """One augmented crop per image (linear-eval training path).  params is
+    [B, 12] in the layout above; image index = row index."""
+    B = images.shape[0]
+    assert params.shape[0] == B
+    out_dtype = out_dtype or images.dtype
+    if use_hip(images):
+        out = ext().two_crop_augment(images.contiguous(),
+                                     params.contiguous(), out_size,
+                                     out_dtype == torch.bfloat16,
+                                     channels_last)
+    else:
+        out = _augment_torch(images, params, out_size).to(out_dtype)
+        if channels_last:
+            out = out.contiguous(memory_format=torch.channels_last)
+    return out
