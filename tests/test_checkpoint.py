@@ -98,3 +98,52 @@ def test_detectron2_converter(tmp_path):
     assert any(k.startswith("res2.") for k in keys)
     assert any(".shortcut." in k or ".shortcut" in k for k in keys)
     assert not any(k.startswith("fc.") for k in keys)
+
+
+def test_resume_matches_continuous_run(tmp_path):
+    """Save at step 2, reload into a fresh model, continue with the same data
+    and step ids: the final state must match the uninterrupted 4-step run
+    (checkpoint completeness: params, EMA state, queue, ptr, momentum)."""
+    import copy
+
+    def make(seed=5):
+        torch.manual_seed(seed)
+        m = moco_amd.MoCo(lambda num_classes: resnet18(num_classes=num_classes),
+                          dim=128, K=64, m=0.99, T=0.2, shuffle_seed=3)
+        o = FusedSGD(m.parameters(), lr=0.01, momentum=0.9, weight_decay=1e-4)
+        return m, o
+
+    torch.manual_seed(77)
+    data = [(torch.randn(4, 3, 32, 32), torch.randn(4, 3, 32, 32))
+            for _ in range(4)]
+
+    def run_steps(model, opt, steps):
+        for s in steps:
+            logits, _ = model(data[s][0], data[s][1], step=s)
+            loss = moco_amd.moco_loss(logits)
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+
+    # continuous 4-step run
+    m1, o1 = make()
+    run_steps(m1, o1, [0, 1, 2, 3])
+
+    # interrupted run: 2 steps -> checkpoint -> fresh objects -> 2 steps
+    m2, o2 = make()
+    run_steps(m2, o2, [0, 1])
+    state = {"state_dict": copy.deepcopy(m2.state_dict()),
+             "optimizer": copy.deepcopy(o2.state_dict())}
+    m3, o3 = make(seed=99)  # different init: everything must come from ckpt
+    m3.load_state_dict(state["state_dict"])
+    o3 = FusedSGD(m3.parameters(), lr=0.01, momentum=0.9, weight_decay=1e-4)
+    o3.load_state_dict(state["optimizer"])
+    run_steps(m3, o3, [2, 3])
+
+    sd1, sd3 = m1.state_dict(), m3.state_dict()
+    for k in sd1:
+        if "num_batches_tracked" in k:
+            assert torch.equal(sd1[k], sd3[k]), k
+        else:
+            assert torch.allclose(sd1[k], sd3[k], atol=1e-5), \
+                (k, (sd1[k] - sd3[k]).abs().max())
