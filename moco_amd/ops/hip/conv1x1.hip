@@ -177,17 +177,34 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     if (DBUF) cur ^= 1;
   }
 
+  // Widened epilogue (guide T21 spirit): the fragment layout leaves each
+  // lane with 2-byte column-scattered values; staging the wave's tile
+  // through LDS turns the store tail into 16-byte coalesced rows.
+  __syncthreads();  // k-loop LDS reads are done; smem is reusable
+  bf16* ct = reinterpret_cast<bf16*>(smem) + wave * wrows * 64;
 #pragma unroll
   for (int mr = 0; mr < MR; ++mr)
 #pragma unroll
     for (int nr = 0; nr < 4; ++nr) {
-      const int col = n0 + wc * 64 + nr * 16 + (lane & 15);
+      const int col = nr * 16 + (lane & 15);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int64_t row = m0 + wr * wrows + mr * 16 + (lane >> 4) * 4 + r;
-        C[row * N + col] = f2bf(acc[mr][nr][r]);
+        const int trow = mr * 16 + (lane >> 4) * 4 + r;
+        ct[trow * 64 + col] = f2bf(acc[mr][nr][r]);
       }
     }
+  __syncthreads();
+  const int npieces = wrows * 64 / 8 / WAVE;  // 16B pieces per lane
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    if (i >= npieces) break;
+    const int piece = i * WAVE + lane;
+    const int trow = piece / 8;
+    const int tcol = (piece % 8) * 8;
+    const int64_t row = m0 + wr * wrows + trow;
+    *reinterpret_cast<bf16x8*>(C + row * N + n0 + wc * 64 + tcol) =
+        *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+  }
 }
 
 }  // namespace

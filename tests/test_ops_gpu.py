@@ -450,3 +450,16 @@ def test_conv1x1_function_matches_conv2d():
             < 0.1 * max(1.0, float(wr.grad.abs().max()))
     finally:
         os.environ.pop("MOCO_AMD_CUSTOM_CONV1X1", None)
+
+
+def test_single_crop_augment_vs_oracle():
+    from moco_amd.ops.augment import (_augment_torch, sample_two_crop_params,
+                                      single_crop_augment)
+    torch.manual_seed(15)
+    B, H, W, S = 3, 64, 64, 32
+    x = torch.rand(B, 3, H, W)
+    params = sample_two_crop_params(B, H, W, S, False)[:B]
+    out = single_crop_augment(x.cuda(), S, params.cuda())
+    ref = _augment_torch(x, params, S)
+    assert out.shape == (B, 3, S, S)
+    assert (out.cpu() - ref).abs().max() < 5e-2
