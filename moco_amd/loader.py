@@ -116,14 +116,19 @@ class ImageFolderDataset(Dataset):
         from PIL import Image
 
         path, label = self.samples[idx]
+        S = self.load_size
         with Image.open(path) as im:
             im = im.convert("RGB")
             w, h = im.size
-            scale = self.load_size / min(w, h)
-            im = im.resize((max(1, round(w * scale)),
-                            max(1, round(h * scale))))
+            # short side -> S, then center crop to SxS: batches must have a
+            # fixed shape for the batched on-GPU augmentation downstream
+            scale = S / min(w, h)
+            nw, nh = max(S, round(w * scale)), max(S, round(h * scale))
+            im = im.resize((nw, nh))
+            left, top = (nw - S) // 2, (nh - S) // 2
+            im = im.crop((left, top, left + S, top + S))
             t = torch.frombuffer(bytearray(im.tobytes()), dtype=torch.uint8)
-            t = t.view(im.size[1], im.size[0], 3).permute(2, 0, 1)
+            t = t.view(S, S, 3).permute(2, 0, 1)
         img = t.float() / 255.0
         if self.transform is not None:
             img = self.transform(img)
