@@ -1,5 +1,6 @@
 from .resnet import (MODEL_REGISTRY, ResNet, build_model, resnet18, resnet34,
-                     resnet50, resnet101, resnet152)
+                     resnet50, resnet101, resnet152, resnext50_32x4d,
+                     resnext101_32x8d, wide_resnet50_2)
 
 __all__ = [
     "ResNet",
@@ -10,4 +11,7 @@ __all__ = [
     "resnet50",
     "resnet101",
     "resnet152",
+    "resnext50_32x4d",
+    "resnext101_32x8d",
+    "wide_resnet50_2",
 ]

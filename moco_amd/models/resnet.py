@@ -23,9 +23,10 @@ from ..ops.conv1x1 import conv1x1 as conv1x1_op
 from ..ops.pool import stem_maxpool
 
 
-def conv3x3(cin: int, cout: int, stride: int = 1, dilation: int = 1) -> nn.Conv2d:
+def conv3x3(cin: int, cout: int, stride: int = 1, dilation: int = 1,
+            groups: int = 1) -> nn.Conv2d:
     return nn.Conv2d(cin, cout, 3, stride=stride, padding=dilation,
-                     dilation=dilation, bias=False)
+                     dilation=dilation, groups=groups, bias=False)
 
 
 def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
@@ -62,13 +63,15 @@ class Bottleneck(nn.Module):
 
     def __init__(self, inplanes: int, planes: int, stride: int = 1,
                  downsample: Optional[nn.Module] = None,
-                 norm_layer: Callable[..., nn.Module] = nn.BatchNorm2d):
+                 norm_layer: Callable[..., nn.Module] = nn.BatchNorm2d,
+                 groups: int = 1, base_width: int = 64):
         super().__init__()
-        self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = norm_layer(planes)
-        self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = norm_layer(planes)
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        width = int(planes * (base_width / 64.0)) * groups
+        self.conv1 = conv1x1(inplanes, width)
+        self.bn1 = norm_layer(width)
+        self.conv2 = conv3x3(width, width, stride, groups=groups)
+        self.bn2 = norm_layer(width)
+        self.conv3 = conv1x1(width, planes * self.expansion)
         self.bn3 = norm_layer(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
@@ -89,9 +92,12 @@ class ResNet(nn.Module):
     def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
                  layers: List[int], num_classes: int = 1000,
                  zero_init_residual: bool = False,
-                 norm_layer: Callable[..., nn.Module] = nn.BatchNorm2d):
+                 norm_layer: Callable[..., nn.Module] = nn.BatchNorm2d,
+                 groups: int = 1, width_per_group: int = 64):
         super().__init__()
         self._norm_layer = norm_layer
+        self.groups = groups
+        self.base_width = width_per_group
         self.inplanes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
         self.bn1 = norm_layer(64)
@@ -128,11 +134,15 @@ class ResNet(nn.Module):
                 conv1x1(self.inplanes, planes * block.expansion, stride),
                 norm_layer(planes * block.expansion),
             )
-        layers = [block(self.inplanes, planes, stride, downsample, norm_layer)]
+        kw = {}
+        if block is Bottleneck:
+            kw = dict(groups=self.groups, base_width=self.base_width)
+        layers = [block(self.inplanes, planes, stride, downsample,
+                        norm_layer, **kw)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
             layers.append(block(self.inplanes, planes,
-                                norm_layer=norm_layer))
+                                norm_layer=norm_layer, **kw))
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -167,12 +177,29 @@ def resnet152(**kwargs) -> ResNet:
     return ResNet(Bottleneck, [3, 8, 36, 3], **kwargs)
 
 
+def resnext50_32x4d(**kwargs) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], groups=32, width_per_group=4,
+                  **kwargs)
+
+
+def resnext101_32x8d(**kwargs) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 23, 3], groups=32, width_per_group=8,
+                  **kwargs)
+
+
+def wide_resnet50_2(**kwargs) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], width_per_group=128, **kwargs)
+
+
 MODEL_REGISTRY = {
     "resnet18": resnet18,
     "resnet34": resnet34,
     "resnet50": resnet50,
     "resnet101": resnet101,
     "resnet152": resnet152,
+    "resnext50_32x4d": resnext50_32x4d,
+    "resnext101_32x8d": resnext101_32x8d,
+    "wide_resnet50_2": wide_resnet50_2,
 }
 
 
