@@ -2,6 +2,8 @@
 32x32 synthetic, queue=1024, world_size 1, no GPU."""
 
 import os
+
+import pytest
 import subprocess
 import sys
 
@@ -66,3 +68,22 @@ def test_v2_flags(tmp_path):
                "--max-steps-per-epoch", "2", "--synthetic-len", "32",
                "--device", "cpu", "--save-dir", str(tmp_path), "-p", "1"])
     assert "mlp=True" in out
+
+
+def test_pretrain_on_imagefolder(tmp_path):
+    """Real-image decode path end to end (PIL ImageFolder -> GPU-style
+    batched aug -> MoCo step)."""
+    pytest.importorskip("PIL")
+    from PIL import Image
+    for cls in ("a", "b"):
+        d = tmp_path / "data" / "train" / cls
+        d.mkdir(parents=True)
+        for i in range(8):
+            Image.new("RGB", (50, 70), color=(i * 20, 50, 100)).save(
+                str(d / f"{i}.png"))
+    out = run(["main_moco.py", str(tmp_path / "data"), "-a", "resnet18",
+               "--moco-k", "32", "--crop-size", "32", "--load-size", "40",
+               "-b", "8", "-j", "0", "--epochs", "1",
+               "--max-steps-per-epoch", "2", "--device", "cpu",
+               "--save-dir", str(tmp_path), "-p", "1"])
+    assert "Epoch: [0][1/2]" in out
