@@ -68,6 +68,7 @@ class MoCo(nn.Module):
 
         self._ema: Optional[MultiTensorEMA] = None
         self._queue_kc: Optional[torch.Tensor] = None  # [K, dim] bf16 (GPU)
+        self._ptr_host: Optional[int] = None  # avoids a per-step device sync
         self._step = 0
 
     # -- key-encoder EMA ----------------------------------------------------
@@ -95,13 +96,18 @@ class MoCo(nn.Module):
     def _load_from_state_dict(self, *args, **kwargs):
         self._queue_kc = None  # rebuilt lazily from the loaded fp32 queue
         self._ema = None
+        self._ptr_host = None
         super()._load_from_state_dict(*args, **kwargs)
 
     @torch.no_grad()
     def _dequeue_and_enqueue(self, keys: torch.Tensor):
         keys = parallel.concat_all_gather(keys)  # C6: all ranks enqueue all
         qkc = self._queue_kc if self.queue.is_cuda else None
-        queue_enqueue_(self.queue, qkc, self.queue_ptr, keys)
+        if self._ptr_host is None:
+            self._ptr_host = int(self.queue_ptr.item())
+        queue_enqueue_(self.queue, qkc, self.queue_ptr, keys,
+                       ptr=self._ptr_host)
+        self._ptr_host = (self._ptr_host + keys.shape[0]) % self.K
 
     # -- shuffling-BN (world-aware; see parallel/comm.py) -------------------
     @torch.no_grad()

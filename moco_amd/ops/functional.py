@@ -162,7 +162,8 @@ def softmax_ce_label0(logits: torch.Tensor) -> torch.Tensor:
 
 @torch.no_grad()
 def queue_enqueue_(queue: torch.Tensor, queue_kc: Optional[torch.Tensor],
-                   queue_ptr: torch.Tensor, keys: torch.Tensor) -> None:
+                   queue_ptr: torch.Tensor, keys: torch.Tensor,
+                   ptr: Optional[int] = None) -> None:
     """Write the gathered key batch into the ring buffer at queue_ptr.
 
     queue:    [C, K] fp32 — canonical, checkpoint-visible layout
@@ -171,11 +172,14 @@ def queue_enqueue_(queue: torch.Tensor, queue_kc: Optional[torch.Tensor],
               or None on the CPU path.
     queue_ptr:[1] int64
     keys:     [B, C] (already gathered across ranks, L2-normalized)
+    ptr:      host-side mirror of queue_ptr — pass it to avoid a per-step
+              device sync (the .item() read)
     """
     B = keys.shape[0]
     K = queue.shape[1]
     assert K % B == 0, f"queue size {K} must be divisible by batch {B}"
-    ptr = int(queue_ptr.item())
+    if ptr is None:
+        ptr = int(queue_ptr.item())
     if use_hip(queue):
         ext().queue_enqueue(queue, queue_kc, keys.contiguous(), ptr)
     else:
