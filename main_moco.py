@@ -85,11 +85,20 @@ def get_parser():
                              "(debug/tests; slower)")
     parser.add_argument("--metrics-file", default="metrics.jsonl", type=str,
                         help="JSONL metrics stream (rank 0; '' disables)")
+    parser.add_argument("--debug-serialize", action="store_true",
+                        help="serialize+sync every kernel launch "
+                             "(AMD_SERIALIZE_KERNEL=3, HIP_LAUNCH_BLOCKING) "
+                             "— the ROCm race-hunting mode; very slow")
     return parser
 
 
 def main(argv=None):
     args = get_parser().parse_args(argv)
+
+    if args.debug_serialize:
+        # must be set before the first HIP call in every worker
+        os.environ["AMD_SERIALIZE_KERNEL"] = "3"
+        os.environ["HIP_LAUNCH_BLOCKING"] = "1"
 
     if args.seed is not None:
         random.seed(args.seed)

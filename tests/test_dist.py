@@ -133,3 +133,11 @@ def test_all_to_all_rows_single_process_identity():
     # world=1: dest_perm semantics -> output slot s = x[argsort(perm)[s]]
     out = parallel.all_to_all_rows(x, perm)
     assert torch.equal(out, x[torch.argsort(perm)])
+
+
+def test_shuffle_exchange_world4():
+    _run_spawn(_worker_shuffle_roundtrip, 29316, world=4)
+
+
+def test_queue_replicated_world4():
+    _run_spawn(_worker_queue_replicated, 29317, world=4)
