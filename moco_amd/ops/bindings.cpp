@@ -278,6 +278,34 @@ torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt) {
   return C;
 }
 
+// Round-2 prototype: GEMM with fused BN-apply (A path) / BN-stats (output).
+std::vector<torch::Tensor> gemm_nt_fused(
+    torch::Tensor A, torch::Tensor Bt, c10::optional<torch::Tensor> a_scale,
+    c10::optional<torch::Tensor> a_shift, bool a_relu, bool want_stats) {
+  CHECK_CUDA_CONTIG(A);
+  CHECK_CUDA_CONTIG(Bt);
+  TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && A.size(1) == Bt.size(1));
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(a_scale.has_value() == a_shift.has_value());
+  if (a_scale.has_value())
+    TORCH_CHECK(a_scale->numel() == K && a_shift->numel() == K);
+  at::cuda::CUDAGuard guard(A.device());
+  auto C = torch::empty({M, N}, A.options());
+  torch::Tensor ssum, ssq;
+  if (want_stats) {
+    ssum = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+    ssq = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+  }
+  CHECK_LAUNCH(launch_gemm_nt_fused_bf16(
+      cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
+      a_scale.has_value() ? a_scale->data_ptr<float>() : nullptr,
+      a_shift.has_value() ? a_shift->data_ptr<float>() : nullptr, a_relu,
+      want_stats ? ssum.data_ptr<float>() : nullptr,
+      want_stats ? ssq.data_ptr<float>() : nullptr));
+  if (want_stats) return {C, ssum, ssq};
+  return {C};
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
@@ -351,5 +379,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_bwd", &bn_bwd);
   mod.def("maxpool_fwd", &maxpool_fwd);
   mod.def("gemm_nt_bf16", &gemm_nt_bf16);
+  mod.def("gemm_nt_fused", &gemm_nt_fused);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

@@ -122,10 +122,20 @@ __device__ __forceinline__ bf16x8 lds_frag(const char* smem, int row,
 // BNT = 128: 2x2 wave grid, 64x64 per wave (MR=NR=4)
 // BNT =  64: 4x1 wave grid, 32x64 per wave (MR=2, NR=4)
 // DBUF: double-buffered staging (off for single-K-tile problems)
-template <int BNT, bool DBUF>
+// APPLY (round-2 conv+BN-fusion prototype): transform every A element with
+//   x' = relu?(x * scale[k] + shift[k])  — i.e. the BatchNorm apply of the
+//   PREVIOUS layer folded into this GEMM's fragment path (scale = rstd*gamma,
+//   shift = beta - mean*rstd*gamma, precomputed per channel on the host).
+// STATS: accumulate per-output-channel (sum, sumsq) of the fp32 accumulator
+//   tile into stats_sum/stats_sumsq (pre-zeroed, fp32 atomics) — the BN
+//   statistics of THIS layer's output for free.
+template <int BNT, bool DBUF, bool APPLY, bool STATS>
 __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
-    bf16* __restrict__ C, int64_t M, int N, int K) {
+    bf16* __restrict__ C, int64_t M, int N, int K,
+    const float* __restrict__ a_scale, const float* __restrict__ a_shift,
+    bool a_relu, float* __restrict__ stats_sum,
+    float* __restrict__ stats_sumsq) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int MR = BNT == 128 ? 4 : 2;
   constexpr int ABYTES = BM * BK * 2;    // 16 KiB
@@ -162,6 +172,25 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
 #pragma unroll
       for (int mr = 0; mr < MR; ++mr)
         af[mr] = lds_frag(abuf, wr * wrows + mr * 16 + (lane & 15), kbyte);
+      if (APPLY) {
+        const int c0 = t * BK + ks * 32 + (lane >> 4) * 8;
+        const float4 sc0 = *reinterpret_cast<const float4*>(a_scale + c0);
+        const float4 sc1 = *reinterpret_cast<const float4*>(a_scale + c0 + 4);
+        const float4 sh0 = *reinterpret_cast<const float4*>(a_shift + c0);
+        const float4 sh1 = *reinterpret_cast<const float4*>(a_shift + c0 + 4);
+        const float scs[8] = {sc0.x, sc0.y, sc0.z, sc0.w,
+                              sc1.x, sc1.y, sc1.z, sc1.w};
+        const float shs[8] = {sh0.x, sh0.y, sh0.z, sh0.w,
+                              sh1.x, sh1.y, sh1.z, sh1.w};
+#pragma unroll
+        for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float v = fmaf((float)af[mr][j], scs[j], shs[j]);
+            if (a_relu) v = fmaxf(v, 0.f);
+            af[mr][j] = (__bf16)v;
+          }
+      }
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr)
         bf[nr] = lds_frag(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
@@ -176,6 +205,33 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     }
     if (DBUF || t + 1 < ntiles) __syncthreads();
     if (DBUF) cur ^= 1;
+  }
+
+  if (STATS) {
+    // per-output-channel partial (sum, sumsq) of the fp32 accumulators.
+    // Column c is held by lanes {l, l+16, l+32, l+48}: butterfly those
+    // four, then one atomic pair per (lane<16, nr).
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      float ssum = 0.f, ssq = 0.f;
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float v = acc[mr][nr][r];
+          ssum += v;
+          ssq += v * v;
+        }
+      ssum += __shfl_xor(ssum, 16, WAVE);
+      ssq += __shfl_xor(ssq, 16, WAVE);
+      ssum += __shfl_xor(ssum, 32, WAVE);
+      ssq += __shfl_xor(ssq, 32, WAVE);
+      if ((lane >> 4) == 0) {
+        const int c = n0 + wc * 64 + nr * 16 + lane;
+        atomicAdd(&stats_sum[c], ssum);
+        atomicAdd(&stats_sumsq[c], ssq);
+      }
+    }
   }
 
   // Widened epilogue (guide T21 spirit): the fragment layout leaves each
@@ -223,12 +279,53 @@ extern "C" int launch_gemm_nt_lds_bf16(void* stream, const void* A,
   const int smem = (16384 + BNT * 128) * (dbuf ? 2 : 1);
   dim3 grid((unsigned)(M / 128), (unsigned)(N / BNT));
 #define GO(BNTv, DB)                                                       \
- hipLaunchKernelGGL(( gemm_nt_lds_kernel<BNTv, DB>), dim3(grid), dim3(256), smem, s,                     \
-      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K)
+ hipLaunchKernelGGL(( gemm_nt_lds_kernel<BNTv, DB, false, false>), dim3(grid), dim3(256), smem, s,       \
+      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,        \
+      nullptr, nullptr, false, nullptr, nullptr)
   if (BNT == 128 && dbuf) GO(128, true);
   else if (BNT == 128) GO(128, false);
   else if (dbuf) GO(64, true);
   else GO(64, false);
+#undef GO
+  HIP_CHECK_LAST();
+  return 0;
+}
+
+// Round-2 prototype entry: GEMM with the BN apply of the previous layer
+// fused into the A-fragment path and/or BN statistics of the output fused
+// into the epilogue.  stats_* must be pre-zeroed [N] fp32 when requested.
+extern "C" int launch_gemm_nt_fused_bf16(void* stream, const void* A,
+                                         const void* Bt, void* C, int64_t M,
+                                         int64_t N, int64_t K,
+                                         const float* a_scale,
+                                         const float* a_shift, bool a_relu,
+                                         float* stats_sum,
+                                         float* stats_sumsq) {
+  if (M % 128 != 0 || K % 64 != 0 || (N % 64 != 0))
+    return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int BNT = (N % 128 == 0) ? 128 : 64;
+  const bool dbuf = (K / 64) > 1;
+  const int smem = (16384 + BNT * 128) * (dbuf ? 2 : 1);
+  const bool apply = a_scale != nullptr;
+  const bool stats = stats_sum != nullptr;
+  dim3 grid((unsigned)(M / 128), (unsigned)(N / BNT));
+#define GO(BNTv, DB, AP, ST)                                               \
+ hipLaunchKernelGGL(( gemm_nt_lds_kernel<BNTv, DB, AP, ST>), dim3(grid), dim3(256), smem, s,             \
+      (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,        \
+      a_scale, a_shift, a_relu, stats_sum, stats_sumsq)
+#define GO2(BNTv, DB)                                                      \
+  do {                                                                     \
+    if (apply && stats) GO(BNTv, DB, true, true);                          \
+    else if (apply) GO(BNTv, DB, true, false);                             \
+    else if (stats) GO(BNTv, DB, false, true);                             \
+    else GO(BNTv, DB, false, false);                                       \
+  } while (0)
+  if (BNT == 128 && dbuf) GO2(128, true);
+  else if (BNT == 128) GO2(128, false);
+  else if (dbuf) GO2(64, true);
+  else GO2(64, false);
+#undef GO2
 #undef GO
   HIP_CHECK_LAST();
   return 0;

@@ -110,3 +110,11 @@ extern "C"
 #endif
 int launch_gemm_nt_lds_bf16(void* stream, const void* A, const void* Bt,
                             void* C, int64_t M, int64_t N, int64_t K);
+#ifdef __cplusplus
+extern "C"
+#endif
+int launch_gemm_nt_fused_bf16(void* stream, const void* A, const void* Bt,
+                              void* C, int64_t M, int64_t N, int64_t K,
+                              const float* a_scale, const float* a_shift,
+                              bool a_relu, float* stats_sum,
+                              float* stats_sumsq);

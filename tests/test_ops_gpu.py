@@ -463,3 +463,47 @@ def test_single_crop_augment_vs_oracle():
     ref = _augment_torch(x, params, S)
     assert out.shape == (B, 3, S, S)
     assert (out.cpu() - ref).abs().max() < 5e-2
+
+
+# ---------------------------------------------------------------------------
+# round-2 prototype: fused GEMM + BN apply/stats (env-gated: experimental)
+# ---------------------------------------------------------------------------
+
+import os as _os
+
+experimental = pytest.mark.skipif(
+    _os.environ.get("MOCO_AMD_EXPERIMENTAL") != "1",
+    reason="round-2 prototype; enable with MOCO_AMD_EXPERIMENTAL=1")
+
+
+@experimental
+@pytest.mark.parametrize("M,N,K", [(1536, 128, 64), (128, 64, 256)])
+def test_gemm_nt_fused_apply_and_stats(M, N, K):
+    from moco_amd import _C
+    torch.manual_seed(16)
+    A = (torch.randn(M, K) * 0.7 + 0.2).cuda().to(torch.bfloat16)
+    Bt = (torch.randn(N, K) * 0.08).cuda().to(torch.bfloat16)
+    scale = (torch.rand(K) * 1.5 + 0.25).cuda()
+    shift = torch.randn(K).cuda() * 0.3
+    C, ssum, ssq = _C.gemm_nt_fused(A, Bt, scale, shift, True, True)
+    # oracle: transform A (with bf16 re-rounding, as the kernel does), GEMM
+    Af = A.float()
+    At = torch.relu(Af * scale[None, :] + shift[None, :]) \
+        .to(torch.bfloat16).float()
+    ref = At @ Bt.float().t()
+    assert (C.float() - ref).abs().max() < 0.06 * max(1.0, K ** 0.5 * 0.05)
+    # stats are over the fp32 accumulator = ref columns
+    assert torch.allclose(ssum.cpu(), ref.sum(0).cpu(), rtol=2e-3, atol=0.5)
+    assert torch.allclose(ssq.cpu(), (ref ** 2).sum(0).cpu(), rtol=2e-3,
+                          atol=1.0)
+
+
+@experimental
+def test_gemm_nt_fused_no_options_matches_plain():
+    from moco_amd import _C
+    torch.manual_seed(17)
+    A = torch.randn(256, 64).cuda().to(torch.bfloat16)
+    Bt = torch.randn(128, 64).cuda().to(torch.bfloat16)
+    (C,) = _C.gemm_nt_fused(A, Bt, None, None, False, False)
+    ref = _C.gemm_nt_bf16(A, Bt)
+    assert torch.equal(C, ref)
