@@ -44,6 +44,12 @@ class _FusedBNAct(torch.autograd.Function):
         return (dx, dgamma, dbeta, None, None, None, None, None, None, dres)
 
 
+def _bn_c_supported(C: int) -> bool:
+    """kernel geometry: Q = C/8 channel octets must divide a 256-thread
+    block (C in {64, 128, 256, 512, 1024, 2048})."""
+    return C % 8 == 0 and 1 <= C // 8 <= 256 and 256 % (C // 8) == 0
+
+
 def bn_act(x: torch.Tensor, bn: torch.nn.BatchNorm2d, relu: bool = True,
            residual: Optional[torch.Tensor] = None) -> torch.Tensor:
     """y = relu?(bn(x) + residual?) with nn.BatchNorm2d semantics
@@ -54,6 +60,7 @@ def bn_act(x: torch.Tensor, bn: torch.nn.BatchNorm2d, relu: bool = True,
 
     use_fused = (x.is_cuda and x.dtype == torch.bfloat16
                  and not force_fallback() and has_ext()
+                 and _bn_c_supported(x.shape[1])
                  and x.is_contiguous(memory_format=torch.channels_last)
                  and (residual is None or
                       (residual.dtype == torch.bfloat16 and

@@ -22,10 +22,14 @@ from . import use_hip, ext
 # L2 normalize (row-wise, dim=1)
 # ---------------------------------------------------------------------------
 
+def _l2norm_shape_ok(x: torch.Tensor) -> bool:
+    return x.shape[1] <= 512  # one wave per row, <= 8 elements per lane
+
+
 class _L2NormalizeRows(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor) -> torch.Tensor:
-        if use_hip(x):
+        if use_hip(x) and _l2norm_shape_ok(x):
             y, inv_norm = ext().l2norm_rows_fwd(x.contiguous())
         else:
             x = x.contiguous()
@@ -39,7 +43,7 @@ class _L2NormalizeRows(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out: torch.Tensor):
         y, inv_norm = ctx.saved_tensors
-        if use_hip(grad_out):
+        if use_hip(grad_out) and _l2norm_shape_ok(grad_out):
             gx = ext().l2norm_rows_bwd(grad_out.contiguous(), y, inv_norm)
         else:
             g = grad_out.float()
@@ -83,7 +87,13 @@ class _ContrastiveLogits(torch.autograd.Function):
                 T: float) -> torch.Tensor:
         q = q.contiguous()
         k = k.contiguous()
-        if use_hip(q):
+        # MFMA kernel constraints (non-standard dims run the hipBLASLt
+        # einsum path on GPU instead — a documented shape fallback, not a
+        # silent downgrade of the supported path)
+        C, K = q.shape[1], queue_kc.shape[0]
+        shapes_ok = (C % 32 == 0 and C <= 512 and K % 32 == 0)
+        ctx.shapes_ok = shapes_ok
+        if use_hip(q) and shapes_ok:
             logits = ext().contrastive_fwd(q, k, queue_kc.contiguous(),
                                            1.0 / T)
         else:
@@ -102,7 +112,7 @@ class _ContrastiveLogits(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_logits: torch.Tensor):
         k, queue_ck = ctx.saved_tensors
-        if use_hip(grad_logits):
+        if use_hip(grad_logits) and ctx.shapes_ok:
             gq = ext().contrastive_bwd(grad_logits.contiguous(), k,
                                        queue_ck.contiguous(), ctx.inv_T)
         else:

@@ -106,3 +106,23 @@ def test_fused_sgd_state_roundtrip():
     b1 = opt.state[p]["momentum_buffer"]
     b2 = opt2.state[p]["momentum_buffer"]
     assert torch.allclose(b1, b2)
+
+
+def test_bn_channel_support_predicate():
+    from moco_amd.ops.bn import _bn_c_supported
+    for c in (64, 128, 256, 512, 1024, 2048):
+        assert _bn_c_supported(c), c
+    for c in (96, 100, 3, 24, 4096, 160):
+        assert not _bn_c_supported(c), c
+
+
+def test_contrastive_odd_dim_cpu_path():
+    # dim not divisible by 32: must still work via the fallback math
+    q = torch.randn(4, 100, requires_grad=True)
+    k = torch.randn(4, 100)
+    queue_kc = torch.randn(16, 100)
+    logits = contrastive_logits(q, k, queue_kc, queue_kc.t().contiguous(),
+                                0.2)
+    assert logits.shape == (4, 17)
+    logits.sum().backward()
+    assert q.grad is not None
