@@ -59,6 +59,8 @@ def get_parser():
     p.add_argument("--multiprocessing-distributed", action="store_true")
     p.add_argument("--pretrained", default="", type=str,
                    help="path to moco pretrained checkpoint")
+    p.add_argument("-e", "--evaluate", action="store_true",
+                   help="evaluate on the validation set and exit")
     p.add_argument("--device", default=None, choices=["cuda", "cpu"])
     p.add_argument("--lincls-dtype", default="fp32", choices=["fp32", "bf16"],
                    help="bf16 autocast for the frozen-backbone forward")
@@ -169,6 +171,9 @@ def main(argv=None):
                               num_workers=args.workers, drop_last=True)
     val_loader = DataLoader(val_set, batch_size=args.batch_size,
                             shuffle=False, num_workers=args.workers)
+
+    if args.evaluate:
+        return validate(val_loader, model, criterion, device, args)
 
     for epoch in range(args.start_epoch, args.epochs):
         if train_sampler is not None:

@@ -87,3 +87,18 @@ def test_pretrain_on_imagefolder(tmp_path):
                "--max-steps-per-epoch", "2", "--device", "cpu",
                "--save-dir", str(tmp_path), "-p", "1"])
     assert "Epoch: [0][1/2]" in out
+
+
+def test_lincls_evaluate_only(tmp_path):
+    run(["main_moco.py", "synthetic", "-a", "resnet18", "--moco-k", "32",
+         "--crop-size", "32", "--load-size", "40", "-b", "8", "-j", "0",
+         "--epochs", "1", "--max-steps-per-epoch", "1",
+         "--synthetic-len", "16", "--device", "cpu",
+         "--save-dir", str(tmp_path), "-p", "1"])
+    out = run(["main_lincls.py", "synthetic", "-a", "resnet18",
+               "--pretrained", str(tmp_path / "checkpoint_0000.pth.tar"),
+               "--device", "cpu", "-b", "8", "-j", "0", "--evaluate",
+               "--synthetic-len", "32", "--crop-size", "32",
+               "--load-size", "40", "--num-classes", "4",
+               "--save-dir", str(tmp_path)])
+    assert "Acc@1" in out
