@@ -51,13 +51,14 @@ class ProgressMeter:
 def accuracy(output: torch.Tensor, target: torch.Tensor, topk=(1,)):
     """Top-k accuracy over the given logits (during pretraining these are the
     contrastive logits: 'did q match k+ against K negatives')."""
-    maxk = max(topk)
+    maxk = min(max(topk), output.size(1))  # fewer classes than k: clamp
     batch_size = target.size(0)
     _, pred = output.float().topk(maxk, 1, True, True)
     pred = pred.t()
     correct = pred.eq(target.view(1, -1).expand_as(pred))
     res = []
     for k in topk:
-        correct_k = correct[:k].reshape(-1).float().sum(0, keepdim=True)
+        kk = min(k, maxk)
+        correct_k = correct[:kk].reshape(-1).float().sum(0, keepdim=True)
         res.append(correct_k.mul_(100.0 / batch_size))
     return res
