@@ -102,3 +102,16 @@ def test_lincls_evaluate_only(tmp_path):
                "--load-size", "40", "--num-classes", "4",
                "--save-dir", str(tmp_path)])
     assert "Acc@1" in out
+
+
+def test_mp_spawn_launch_path(tmp_path):
+    """The reference-style mp.spawn launcher (--multiprocessing-distributed)
+    on one CPU 'device'."""
+    out = run(["main_moco.py", "synthetic", "-a", "resnet18",
+               "--moco-k", "32", "--crop-size", "32", "--load-size", "40",
+               "-b", "8", "-j", "0", "--epochs", "1",
+               "--max-steps-per-epoch", "1", "--synthetic-len", "16",
+               "--device", "cpu", "--save-dir", str(tmp_path), "-p", "1",
+               "--multiprocessing-distributed", "--dist-backend", "gloo",
+               "--dist-url", "tcp://127.0.0.1:29411"])
+    assert "Epoch: [0]" in out
