@@ -147,7 +147,10 @@ class MoCo(nn.Module):
             k = l2_normalize(k, dim=1)
             k = self._batch_unshuffle(k, idx_unshuffle)
 
-        if q.is_cuda:
+        if q.is_cuda and (q.dtype == torch.bfloat16
+                          or torch.is_autocast_enabled()):
+            # bf16 MFMA path; an explicit fp32 run keeps fp32 precision and
+            # takes the hipBLASLt fallback inside contrastive_logits
             q = q.to(torch.bfloat16)
             k = k.to(torch.bfloat16)
         logits = contrastive_logits(q, k, self._working_queue(), self.queue,

@@ -91,7 +91,8 @@ class _ContrastiveLogits(torch.autograd.Function):
         # einsum path on GPU instead — a documented shape fallback, not a
         # silent downgrade of the supported path)
         C, K = q.shape[1], queue_kc.shape[0]
-        shapes_ok = (C % 32 == 0 and C <= 512 and K % 32 == 0)
+        shapes_ok = (C % 32 == 0 and C <= 512 and K % 32 == 0
+                     and q.dtype == torch.bfloat16)
         ctx.shapes_ok = shapes_ok
         if use_hip(q) and shapes_ok:
             logits = ext().contrastive_fwd(q, k, queue_kc.contiguous(),
