@@ -306,6 +306,47 @@ std::vector<torch::Tensor> gemm_nt_fused(
   return {C};
 }
 
+// Round-2 prototype: 3x3 pad-1 conv (stride 1/2) as implicit MFMA GEMM,
+// with optional fused BN-apply (input path) / BN-stats (output).
+std::vector<torch::Tensor> conv3x3_fwd(
+    torch::Tensor x, torch::Tensor w, int64_t stride,
+    c10::optional<torch::Tensor> a_scale, c10::optional<torch::Tensor> a_shift,
+    bool a_relu, bool want_stats, c10::optional<torch::Tensor> neutral) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x) && x.dim() == 4);
+  TORCH_CHECK(is_bf16(w) && w.dim() == 4 && w.size(2) == 3 && w.size(3) == 3);
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast) &&
+              w.is_contiguous(torch::MemoryFormat::ChannelsLast),
+              "conv3x3_fwd wants channels_last x and w");
+  const int64_t N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int64_t Co = w.size(0);
+  TORCH_CHECK(w.size(1) == Ci);
+  const int64_t Ho = (H - 1) / stride + 1, Wo = (W - 1) / stride + 1;
+  at::cuda::CUDAGuard guard(x.device());
+  auto out = torch::empty({N, Co, Ho, Wo}, x.options())
+                 .contiguous(torch::MemoryFormat::ChannelsLast);
+  torch::Tensor zp;
+  if (neutral.has_value()) {
+    TORCH_CHECK(is_bf16(*neutral) && neutral->numel() >= Ci + 8);
+    zp = *neutral;
+  } else {
+    zp = torch::zeros({Ci + 8}, x.options());
+  }
+  torch::Tensor ssum, ssq;
+  if (want_stats) {
+    ssum = torch::zeros({Co}, x.options().dtype(torch::kFloat32));
+    ssq = torch::zeros({Co}, x.options().dtype(torch::kFloat32));
+  }
+  CHECK_LAUNCH(launch_conv3x3_fwd(
+      cur_stream(), x.data_ptr(), w.data_ptr(), zp.data_ptr(),
+      out.data_ptr(), N, H, W, Ci, Co, stride,
+      a_scale.has_value() ? a_scale->data_ptr<float>() : nullptr,
+      a_shift.has_value() ? a_shift->data_ptr<float>() : nullptr, a_relu,
+      want_stats ? ssum.data_ptr<float>() : nullptr,
+      want_stats ? ssq.data_ptr<float>() : nullptr));
+  if (want_stats) return {out, ssum, ssq};
+  return {out};
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
@@ -380,5 +421,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("maxpool_fwd", &maxpool_fwd);
   mod.def("gemm_nt_bf16", &gemm_nt_bf16);
   mod.def("gemm_nt_fused", &gemm_nt_fused);
+  mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

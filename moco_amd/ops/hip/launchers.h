@@ -118,3 +118,13 @@ int launch_gemm_nt_fused_bf16(void* stream, const void* A, const void* Bt,
                               const float* a_scale, const float* a_shift,
                               bool a_relu, float* stats_sum,
                               float* stats_sumsq);
+#ifdef __cplusplus
+extern "C"
+#endif
+// 3x3 pad-1 conv (stride 1/2) as implicit MFMA GEMM (round-2 prototype).
+// zero_page: [Ci + 8] bf16 — zeros, or the BN-apply neutral values.
+int launch_conv3x3_fwd(void* stream, const void* x, const void* w,
+                       const void* zero_page, void* out, int64_t N, int64_t H,
+                       int64_t W, int64_t Ci, int64_t Co, int64_t stride,
+                       const float* a_scale, const float* a_shift,
+                       bool a_relu, float* stats_sum, float* stats_sumsq);

@@ -26,6 +26,7 @@ sources = [
     "moco_amd/ops/hip/bn.hip",
     "moco_amd/ops/hip/pool.hip",
     "moco_amd/ops/hip/conv1x1.hip",
+    "moco_amd/ops/hip/conv3x3.hip",
     "moco_amd/ops/hip/augment.hip",
 ]
 

@@ -507,3 +507,56 @@ def test_gemm_nt_fused_no_options_matches_plain():
     (C,) = _C.gemm_nt_fused(A, Bt, None, None, False, False)
     ref = _C.gemm_nt_bf16(A, Bt)
     assert torch.equal(C, ref)
+
+
+@experimental
+@pytest.mark.parametrize("N,Ci,H,Co,stride", [
+    (8, 64, 16, 64, 1),     # M = 2048
+    (8, 64, 16, 128, 1),
+    (2, 128, 32, 64, 2),    # stride-2: Ho=16, M = 512
+    (8, 256, 8, 192, 1),    # Co = 192 -> BNT 64 path
+])
+def test_conv3x3_fwd_vs_conv2d(N, Ci, H, Co, stride):
+    from moco_amd import _C
+    torch.manual_seed(18)
+    x = (torch.randn(N, Ci, H, H) * 0.5).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(Co, Ci, 3, 3) * 0.05).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    (out,) = _C.conv3x3_fwd(x, w, stride, None, None, False, False, None)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), stride=stride,
+                                     padding=1)
+    assert out.shape == ref.shape
+    tol = 0.05 * max(1.0, (9 * Ci) ** 0.5 * 0.025)
+    assert (out.float() - ref).abs().max() < tol, \
+        (out.float() - ref).abs().max()
+
+
+@experimental
+def test_conv3x3_fused_apply_stats():
+    """conv3x3 with the previous BN's apply folded in (neutral-page padding)
+    + output stats."""
+    from moco_amd import _C
+    torch.manual_seed(19)
+    N, Ci, H, Co = 8, 64, 16, 64
+    x = (torch.randn(N, Ci, H, H)).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(Co, Ci, 3, 3) * 0.05).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    scale = (torch.rand(Ci) + 0.5).cuda()
+    shift = (torch.randn(Ci) * 0.2).cuda()
+    neutral = torch.zeros(Ci + 8, dtype=torch.bfloat16, device="cuda")
+    neutral[:Ci] = (-shift / scale).to(torch.bfloat16)
+    out, ssum, ssq = _C.conv3x3_fwd(x, w, 1, scale, shift, True, True,
+                                    neutral)
+    # oracle: transform x (bf16 re-round), THEN zero-pad, then conv
+    xt = torch.relu(x.float() * scale.view(1, -1, 1, 1)
+                    + shift.view(1, -1, 1, 1)).to(torch.bfloat16).float()
+    ref = torch.nn.functional.conv2d(xt, w.float(), padding=1)
+    d = (out.float() - ref).abs()
+    # neutral-page padding quantizes -shift/scale to bf16: border columns
+    # carry a small extra error, bounded here
+    assert d.max() < 0.15, d.max()
+    assert d[:, :, 1:-1, 1:-1].max() < 0.06, d[:, :, 1:-1, 1:-1].max()
+    assert torch.allclose(ssum.cpu(), ref.sum((0, 2, 3)).cpu(), rtol=5e-3,
+                          atol=1.0)
