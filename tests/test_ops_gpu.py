@@ -560,3 +560,21 @@ def test_conv3x3_fused_apply_stats():
     assert d[:, :, 1:-1, 1:-1].max() < 0.06, d[:, :, 1:-1, 1:-1].max()
     assert torch.allclose(ssum.cpu(), ref.sum((0, 2, 3)).cpu(), rtol=5e-3,
                           atol=1.0)
+
+
+@experimental
+def test_conv3x3_dgrad_s1_matches_autograd():
+    from moco_amd.ops.conv3x3 import conv3x3_dgrad_s1
+    torch.manual_seed(20)
+    N, Ci, H, Co = 8, 64, 16, 64
+    x = (torch.randn(N, Ci, H, H) * 0.5).cuda().float().requires_grad_(True)
+    w = (torch.randn(Co, Ci, 3, 3) * 0.05).cuda().float()
+    y = torch.nn.functional.conv2d(x, w, padding=1)
+    g = torch.randn_like(y)
+    y.backward(g)
+    dx = conv3x3_dgrad_s1(
+        g.to(torch.bfloat16).contiguous(memory_format=torch.channels_last),
+        w.to(torch.bfloat16).contiguous(memory_format=torch.channels_last))
+    tol = 0.05 * max(1.0, (9 * Co) ** 0.5 * 0.025)
+    assert (dx.float() - x.grad).abs().max() < tol, \
+        (dx.float() - x.grad).abs().max()
