@@ -347,6 +347,25 @@ std::vector<torch::Tensor> conv3x3_fwd(
   return {out};
 }
 
+torch::Tensor conv1x1_strided(torch::Tensor x, torch::Tensor w,
+                              int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x) && x.dim() == 4);
+  TORCH_CHECK(is_bf16(w) && w.dim() == 4 && w.size(2) == 1 && w.size(3) == 1);
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast));
+  const int64_t N = x.size(0), Ci = x.size(1), H = x.size(2), W = x.size(3);
+  const int64_t Co = w.size(0);
+  TORCH_CHECK(w.size(1) == Ci);
+  const int64_t Ho = (H - 1) / stride + 1, Wo = (W - 1) / stride + 1;
+  at::cuda::CUDAGuard guard(x.device());
+  auto out = torch::empty({N, Co, Ho, Wo}, x.options())
+                 .contiguous(torch::MemoryFormat::ChannelsLast);
+  CHECK_LAUNCH(launch_conv1x1_strided(cur_stream(), x.data_ptr(),
+                                      w.contiguous().data_ptr(),
+                                      out.data_ptr(), N, H, W, Ci, Co,
+                                      stride));
+  return out;
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
@@ -422,5 +441,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nt_bf16", &gemm_nt_bf16);
   mod.def("gemm_nt_fused", &gemm_nt_fused);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
+  mod.def("conv1x1_strided", &conv1x1_strided);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

@@ -271,3 +271,144 @@ extern "C" int launch_conv3x3_fwd(void* stream, const void* x, const void* w,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Strided 1x1 conv (the bottleneck downsample path): the same implicit-GEMM
+// structure with a trivial row map — output row (n, ho, wo) reads input row
+// (n, ho*s, wo*s), always in bounds, no padding page needed.
+// ---------------------------------------------------------------------------
+namespace {
+
+__device__ __forceinline__ void stage_strided_rows(
+    const bf16* __restrict__ x, char* smem, int wave, int lane, int64_t m0,
+    int kt, int H, int W, int Ci, int Ho, int Wo, int stride) {
+  const int ci_b = kt * BK3;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int piece = c * 256 + wave * 64 + lane;
+    const int row = piece / 8;
+    const int kb = (piece % 8) * 16;
+    const int ci = ci_b + swz3(row, kb) / 2;
+    const int64_t m = m0 + row;
+    const int wo = (int)(m % Wo);
+    const int64_t t = m / Wo;
+    const int ho = (int)(t % Ho);
+    const int n = (int)(t / Ho);
+    const bf16* src =
+        x + (((int64_t)n * H + ho * stride) * W + wo * stride) * Ci + ci;
+    char* lds_base = smem + (int64_t)(c * 256 + wave * 64) * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)lds_base, 16, 0, 0);
+  }
+}
+
+template <int BNT>
+__global__ __launch_bounds__(256, 2) void conv1x1s_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ Wt,
+    bf16* __restrict__ C, int H, int W, int Ci, int Co, int Ho, int Wo,
+    int stride) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int MR = BNT == 128 ? 4 : 2;
+  constexpr int ABYTES = BM3 * BK3 * 2;
+  constexpr int BBYTES = BNT * BK3 * 2;
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int wr = BNT == 128 ? (wave >> 1) : wave;
+  const int wc = BNT == 128 ? (wave & 1) : 0;
+  const int wrows = BNT == 128 ? 64 : 32;
+  const int64_t m0 = (int64_t)blockIdx.x * BM3;
+  const int n0 = blockIdx.y * BNT;
+
+  f32x4 acc[MR][4] = {};
+  stage_strided_rows(x, smem, wave, lane, m0, 0, H, W, Ci, Ho, Wo, stride);
+  stage_w<BNT>(Wt + (int64_t)n0 * Ci, Ci, smem + ABYTES, wave, lane);
+  __syncthreads();
+
+  int cur = 0;
+  const int ntiles = Ci / BK3;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      char* nbuf = smem + (cur ^ 1) * (ABYTES + BBYTES);
+      stage_strided_rows(x, nbuf, wave, lane, m0, t + 1, H, W, Ci, Ho, Wo,
+                         stride);
+      stage_w<BNT>(Wt + (int64_t)n0 * Ci + (t + 1) * BK3, Ci, nbuf + ABYTES,
+                   wave, lane);
+    }
+    const char* abuf = smem + cur * (ABYTES + BBYTES);
+    const char* bbuf = abuf + ABYTES;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[MR], bf[4];
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+        af[mr] = lds_frag3(abuf, wr * wrows + mr * 16 + (lane & 15), kbyte);
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr)
+        bf[nr] = lds_frag3(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr)
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mr], bf[nr], acc[mr][nr], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  __syncthreads();
+  bf16* ct = reinterpret_cast<bf16*>(smem) + wave * wrows * 64;
+#pragma unroll
+  for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      const int col = nr * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int trow = mr * 16 + (lane >> 4) * 4 + r;
+        ct[trow * 64 + col] = f2bf(acc[mr][nr][r]);
+      }
+    }
+  __syncthreads();
+  const int npieces = wrows * 64 / 8 / WAVE;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    if (i >= npieces) break;
+    const int piece = i * WAVE + lane;
+    const int trow = piece / 8;
+    const int tcol = (piece % 8) * 8;
+    const int64_t row = m0 + wr * wrows + trow;
+    *reinterpret_cast<bf16x8*>(C + row * Co + n0 + wc * 64 + tcol) =
+        *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+  }
+}
+
+}  // namespace
+
+extern "C" int launch_conv1x1_strided(void* stream, const void* x,
+                                      const void* w, void* out, int64_t N,
+                                      int64_t H, int64_t W, int64_t Ci,
+                                      int64_t Co, int64_t stride) {
+  const int64_t Ho = (H - 1) / stride + 1, Wo = (W - 1) / stride + 1;
+  const int64_t M = N * Ho * Wo;
+  if (Ci % 64 != 0 || M % 128 != 0 || Co % 64 != 0)
+    return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int BNT = (Co % 128 == 0) ? 128 : 64;
+  const int smem = (16384 + BNT * 128) * 2;
+  dim3 grid((unsigned)(M / 128), (unsigned)(Co / BNT));
+  if (BNT == 128)
+    conv1x1s_kernel<128><<<grid, 256, smem, s>>>(
+        (const bf16*)x, (const bf16*)w, (bf16*)out, (int)H, (int)W, (int)Ci,
+        (int)Co, (int)Ho, (int)Wo, (int)stride);
+  else
+    conv1x1s_kernel<64><<<grid, 256, smem, s>>>(
+        (const bf16*)x, (const bf16*)w, (bf16*)out, (int)H, (int)W, (int)Ci,
+        (int)Co, (int)Ho, (int)Wo, (int)stride);
+  HIP_CHECK_LAST();
+  return 0;
+}

@@ -128,3 +128,10 @@ int launch_conv3x3_fwd(void* stream, const void* x, const void* w,
                        int64_t W, int64_t Ci, int64_t Co, int64_t stride,
                        const float* a_scale, const float* a_shift,
                        bool a_relu, float* stats_sum, float* stats_sumsq);
+#ifdef __cplusplus
+extern "C"
+#endif
+// strided 1x1 conv (downsample path), NHWC bf16 (round-2 prototype)
+int launch_conv1x1_strided(void* stream, const void* x, const void* w,
+                           void* out, int64_t N, int64_t H, int64_t W,
+                           int64_t Ci, int64_t Co, int64_t stride);

@@ -578,3 +578,20 @@ def test_conv3x3_dgrad_s1_matches_autograd():
     tol = 0.05 * max(1.0, (9 * Co) ** 0.5 * 0.025)
     assert (dx.float() - x.grad).abs().max() < tol, \
         (dx.float() - x.grad).abs().max()
+
+
+@experimental
+@pytest.mark.parametrize("N,Ci,H,Co,stride", [
+    (8, 256, 16, 512, 2), (2, 64, 32, 128, 2), (8, 64, 16, 64, 1)])
+def test_conv1x1_strided_vs_conv2d(N, Ci, H, Co, stride):
+    from moco_amd import _C
+    torch.manual_seed(21)
+    x = (torch.randn(N, Ci, H, H) * 0.5).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(Co, Ci, 1, 1) * 0.05).cuda().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    out = _C.conv1x1_strided(x, w, stride)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), stride=stride)
+    assert out.shape == ref.shape
+    assert (out.float() - ref).abs().max() < 0.05 * max(1.0,
+                                                        Ci ** 0.5 * 0.025)
