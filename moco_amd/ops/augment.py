@@ -122,13 +122,17 @@ def two_crop_augment(images: torch.Tensor, out_size: int = 224,
         params = sample_two_crop_params(B, H, W, out_size, aug_plus,
                                         generator=generator,
                                         device=str(images.device))
-    out_dtype = out_dtype or images.dtype
+    if out_dtype is None:
+        out_dtype = (images.dtype if images.dtype != torch.uint8
+                     else torch.float32)
     if use_hip(images):
         out = ext().two_crop_augment(images.contiguous(), params.contiguous(),
                                      out_size,
                                      out_dtype == torch.bfloat16,
                                      channels_last)
     else:
+        if images.dtype == torch.uint8:
+            images = images.float() / 255.0
         out = _augment_torch(images, params, out_size).to(out_dtype)
         if channels_last:
             out = out.contiguous(memory_format=torch.channels_last)
@@ -287,13 +291,17 @@ def single_crop_augment(images: torch.Tensor, out_size: int,
     [B, 12] in the layout above; image index = row index."""
     B = images.shape[0]
     assert params.shape[0] == B
-    out_dtype = out_dtype or images.dtype
+    if out_dtype is None:
+        out_dtype = (images.dtype if images.dtype != torch.uint8
+                     else torch.float32)
     if use_hip(images):
         out = ext().two_crop_augment(images.contiguous(),
                                      params.contiguous(), out_size,
                                      out_dtype == torch.bfloat16,
                                      channels_last)
     else:
+        if images.dtype == torch.uint8:
+            images = images.float() / 255.0
         out = _augment_torch(images, params, out_size).to(out_dtype)
         if channels_last:
             out = out.contiguous(memory_format=torch.channels_last)

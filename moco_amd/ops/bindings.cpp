@@ -398,8 +398,10 @@ torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
                                int64_t S, bool out_bf16, bool channels_last) {
   CHECK_CUDA_CONTIG(src);
   CHECK_CUDA_CONTIG(params);
-  TORCH_CHECK(src.scalar_type() == torch::kFloat32 && src.dim() == 4 &&
-              src.size(1) == 3, "src must be fp32 [B,3,H,W]");
+  TORCH_CHECK((src.scalar_type() == torch::kFloat32 ||
+               src.scalar_type() == torch::kUInt8) &&
+              src.dim() == 4 && src.size(1) == 3,
+              "src must be fp32 or uint8 [B,3,H,W]");
   TORCH_CHECK(params.scalar_type() == torch::kFloat32 &&
               params.size(1) == 12);
   const int64_t B = src.size(0), H = src.size(2), W = src.size(3);
@@ -414,7 +416,8 @@ torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
   auto tmp2 = torch::empty({M, 3, S, S}, f32);
   auto graysum = torch::zeros({M}, f32);
   CHECK_LAUNCH(launch_two_crop_augment(
-      cur_stream(), src.data_ptr<float>(), params.data_ptr<float>(),
+      cur_stream(), src.data_ptr(),
+      src.scalar_type() == torch::kUInt8, params.data_ptr<float>(),
       out.data_ptr(), tmp1.data_ptr<float>(), tmp2.data_ptr<float>(),
       graysum.data_ptr<float>(), B, M, H, W, S, out_bf16, channels_last));
   return out;

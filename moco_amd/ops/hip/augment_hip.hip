@@ -34,8 +34,20 @@ __device__ __forceinline__ float gray_of(float r, float g, float b) {
 }
 
 // ---- pass 1: crop-resize + gray mean --------------------------------------
+// ST: float (already [0,1]) or unsigned char (decoded bytes, scaled /255 —
+// real-image pipelines ship uint8 and save 4x loader/H2D bandwidth)
+template <typename ST>
+__device__ __forceinline__ float src_val(ST v);
+template <>
+__device__ __forceinline__ float src_val<float>(float v) { return v; }
+template <>
+__device__ __forceinline__ float src_val<unsigned char>(unsigned char v) {
+  return (float)v * (1.f / 255.f);
+}
+
+template <typename ST>
 __global__ __launch_bounds__(256) void aug_crop_resize_kernel(
-    const float* __restrict__ src, const float* __restrict__ params,
+    const ST* __restrict__ src, const float* __restrict__ params,
     float* __restrict__ tmp1, float* __restrict__ graysum, int B, int H,
     int W, int S) {
   const int m = blockIdx.y;
@@ -53,15 +65,15 @@ __global__ __launch_bounds__(256) void aug_crop_resize_kernel(
     const int x1 = min(max((int)fx0 + 1, 0), W - 1);
     const int y0 = min(max((int)fy0, 0), H - 1);
     const int y1 = min(max((int)fy0 + 1, 0), H - 1);
-    const float* sb = src + img * 3 * (int64_t)H * W;
+    const ST* sb = src + img * 3 * (int64_t)H * W;
     float out3[3];
 #pragma unroll
     for (int ch = 0; ch < 3; ++ch) {
-      const float* sc = sb + (int64_t)ch * H * W;
-      const float v00 = sc[(int64_t)y0 * W + x0];
-      const float v01 = sc[(int64_t)y0 * W + x1];
-      const float v10 = sc[(int64_t)y1 * W + x0];
-      const float v11 = sc[(int64_t)y1 * W + x1];
+      const ST* sc = sb + (int64_t)ch * H * W;
+      const float v00 = src_val<ST>(sc[(int64_t)y0 * W + x0]);
+      const float v01 = src_val<ST>(sc[(int64_t)y0 * W + x1]);
+      const float v10 = src_val<ST>(sc[(int64_t)y1 * W + x0]);
+      const float v11 = src_val<ST>(sc[(int64_t)y1 * W + x1]);
       const float top = v00 * (1.f - wx) + v01 * wx;
       const float bot = v10 * (1.f - wx) + v11 * wx;
       const float v = top * (1.f - wy) + bot * wy;
@@ -252,16 +264,22 @@ __global__ __launch_bounds__(256) void aug_blur_v_finalize_kernel(
 
 }  // namespace
 
-extern "C" int launch_two_crop_augment(void* stream, const float* src,
-                                       const float* params, void* out,
-                                       float* tmp1, float* tmp2,
+extern "C" int launch_two_crop_augment(void* stream, const void* src,
+                                       bool src_u8, const float* params,
+                                       void* out, float* tmp1, float* tmp2,
                                        float* graysum, int64_t B, int64_t M,
                                        int64_t H, int64_t W, int64_t S,
                                        bool out_bf16, bool channels_last) {
   hipStream_t s = (hipStream_t)stream;
   const dim3 grid((unsigned)((S * S + 255) / 256), (unsigned)M);
- hipLaunchKernelGGL(( aug_crop_resize_kernel), dim3(grid), dim3(256), 0, s, src, params, tmp1, graysum,
-                                              (int)B, (int)H, (int)W, (int)S);
+  if (src_u8)
+   hipLaunchKernelGGL(( aug_crop_resize_kernel<unsigned char>), dim3(grid), dim3(256), 0, s, 
+        (const unsigned char*)src, params, tmp1, graysum, (int)B, (int)H,
+        (int)W, (int)S);
+  else
+   hipLaunchKernelGGL(( aug_crop_resize_kernel<float>), dim3(grid), dim3(256), 0, s, 
+        (const float*)src, params, tmp1, graysum, (int)B, (int)H, (int)W,
+        (int)S);
   HIP_CHECK_LAST();
  hipLaunchKernelGGL(( aug_jitter_kernel), dim3(grid), dim3(256), 0, s, tmp1, params, graysum, tmp2,
                                          (int)S);

@@ -75,7 +75,7 @@ int launch_bn_bwd_dx(void* stream, const void* g, const void* y,
 // src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,
 // NCHW or NHWC per flags); tmp1/tmp2: [M,3,S,S] fp32 workspaces;
 // graysum: [M] fp32 (pre-zeroed)
-int launch_two_crop_augment(void* stream, const float* src,
+int launch_two_crop_augment(void* stream, const void* src, bool src_u8,
                             const float* params, void* out, float* tmp1,
                             float* tmp2, float* graysum, int64_t B, int64_t M,
                             int64_t H, int64_t W, int64_t S, bool out_bf16,

@@ -122,3 +122,19 @@ def test_two_crop_deterministic_given_params():
     q2, k2 = two_crop_augment(x, 32, True, params=params,
                               channels_last=False)
     assert torch.equal(q1, q2) and torch.equal(k1, k2)
+
+
+def test_uint8_source_cpu_equivalence():
+    """uint8 sources must match the fp32/255 path exactly (the GPU kernel
+    scales in pass 1; the CPU fallback converts up front)."""
+    from moco_amd.ops.augment import two_crop_augment, sample_two_crop_params
+    torch.manual_seed(3)
+    u8 = (torch.rand(2, 3, 48, 48) * 255).to(torch.uint8)
+    params = sample_two_crop_params(2, 48, 48, 32, True)
+    q1, k1 = two_crop_augment(u8, 32, True, params=params,
+                              channels_last=False)
+    q2, k2 = two_crop_augment(u8.float() / 255.0, 32, True, params=params,
+                              channels_last=False)
+    assert q1.dtype == torch.float32
+    assert torch.allclose(q1, q2, atol=1e-6)
+    assert torch.allclose(k1, k2, atol=1e-6)

@@ -595,3 +595,15 @@ def test_conv1x1_strided_vs_conv2d(N, Ci, H, Co, stride):
     assert out.shape == ref.shape
     assert (out.float() - ref).abs().max() < 0.05 * max(1.0,
                                                         Ci ** 0.5 * 0.025)
+
+
+def test_two_crop_augment_uint8_gpu():
+    from moco_amd import _C
+    from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
+    torch.manual_seed(22)
+    B, H, W, S = 3, 64, 64, 32
+    u8 = (torch.rand(B, 3, H, W) * 255).to(torch.uint8)
+    params = sample_two_crop_params(B, H, W, S, True)
+    out = _C.two_crop_augment(u8.cuda(), params.cuda(), S, False, False)
+    ref = _augment_torch(u8.float() / 255.0, params, S)
+    assert (out.cpu() - ref).abs().max() < 5e-2
