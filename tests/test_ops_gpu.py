@@ -597,6 +597,7 @@ def test_conv1x1_strided_vs_conv2d(N, Ci, H, Co, stride):
                                                         Ci ** 0.5 * 0.025)
 
 
+@experimental
 def test_two_crop_augment_uint8_gpu():
     from moco_amd import _C
     from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
