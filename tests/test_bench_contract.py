@@ -27,3 +27,20 @@ def test_bench_json_contract():
     assert out["data"] == "synthetic"
     assert out["value"] > 0
     assert "global_batch" in out["config"] and "parallelism" in out["config"]
+
+
+def test_bench_under_torchrun_world2():
+    """The exact driver launch shape for N>1 (CPU/gloo fallback here)."""
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29733", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1"],
+        cwd=REPO, timeout=600, capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr[-2000:]
+    json_lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(json_lines) == 1  # rank 0 only
+    out = json.loads(json_lines[0])
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["config"]["global_batch"] == 16  # cpu debug batch 8 x 2
