@@ -366,6 +366,30 @@ torch::Tensor conv1x1_strided(torch::Tensor x, torch::Tensor w,
   return out;
 }
 
+// Round-2 prototype: dgrad GEMM + fused BN-backward epilogue.
+std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
+                                             torch::Tensor Bt,
+                                             torch::Tensor y, torch::Tensor z,
+                                             torch::Tensor mean,
+                                             torch::Tensor rstd) {
+  CHECK_CUDA_CONTIG(A);
+  CHECK_CUDA_CONTIG(Bt);
+  TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && is_bf16(y) && is_bf16(z));
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(y.numel() == M * N && z.numel() == M * N);
+  TORCH_CHECK(mean.numel() == N && rstd.numel() == N);
+  at::cuda::CUDAGuard guard(A.device());
+  auto C = torch::empty({M, N}, A.options());
+  auto dbeta = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+  auto dgamma = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+  CHECK_LAUNCH(launch_gemm_nt_dgrad_epi(
+      cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
+      y.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
+      rstd.data_ptr<float>(), dbeta.data_ptr<float>(),
+      dgamma.data_ptr<float>()));
+  return {C, dbeta, dgamma};
+}
+
 // ---------------------------------------------------------------------------
 // MaxPool2d 3x3/s2/p1 (ResNet stem), NHWC bf16
 std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
@@ -445,5 +469,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nt_fused", &gemm_nt_fused);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv1x1_strided", &conv1x1_strided);
+  mod.def("gemm_nt_dgrad_epi", &gemm_nt_dgrad_epi);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

@@ -329,3 +329,170 @@ extern "C" int launch_gemm_nt_fused_bf16(void* stream, const void* A,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Round-2 prototype: dgrad GEMM with the NEXT stage's BN backward folded
+// into the epilogue.  For the chain y = relu(bn(z) + res), the consumer
+// conv's dgrad GEMM produces dy; this variant post-processes each output
+// tile in-register/LDS:
+//     g'      = dy * (y > 0)                      (the relu mask)
+//     dbeta  += sum g'          per channel       (fp32 atomics)
+//     dgamma += sum g' * zhat,  zhat = (z-mean)*rstd
+// and stores g' — replacing the separate 3-read bn_bwd_reduce pass.
+// Channel bookkeeping rides the widened LDS-transpose epilogue: a lane's
+// 16-byte piece always covers the same 8 output channels ((lane%8)*8), so
+// per-channel partials accumulate in 8 registers and reduce across the 8
+// same-class lanes with three butterflies.
+// ---------------------------------------------------------------------------
+namespace {
+
+template <int BNT>
+__global__ __launch_bounds__(256, 2) void gemm_nt_dgrad_epi_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ Bt,
+    bf16* __restrict__ C, int64_t M, int N, int K,
+    const bf16* __restrict__ y, const bf16* __restrict__ z,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ dbeta, float* __restrict__ dgamma) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int MR = BNT == 128 ? 4 : 2;
+  constexpr int ABYTES = BM * BK * 2;
+  constexpr int BBYTES = BNT * BK * 2;
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int wr = BNT == 128 ? (wave >> 1) : wave;
+  const int wc = BNT == 128 ? (wave & 1) : 0;
+  const int wrows = BNT == 128 ? 64 : 32;
+  const int64_t m0 = (int64_t)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BNT;
+
+  f32x4 acc[MR][4] = {};
+  stage_tile<BM>(A + m0 * K, K, smem, wave, lane);
+  stage_tile<BNT>(Bt + (int64_t)n0 * K, K, smem + ABYTES, wave, lane);
+  __syncthreads();
+
+  int cur = 0;
+  const int ntiles = K / BK;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      char* nbuf = smem + (cur ^ 1) * (ABYTES + BBYTES);
+      stage_tile<BM>(A + m0 * K + (t + 1) * BK, K, nbuf, wave, lane);
+      stage_tile<BNT>(Bt + (int64_t)n0 * K + (t + 1) * BK, K, nbuf + ABYTES,
+                      wave, lane);
+    }
+    const char* abuf = smem + cur * (ABYTES + BBYTES);
+    const char* bbuf = abuf + ABYTES;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[MR], bf[4];
+      const int kbyte = ks * 64 + (lane >> 4) * 16;
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+        af[mr] = lds_frag(abuf, wr * wrows + mr * 16 + (lane & 15), kbyte);
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr)
+        bf[nr] = lds_frag(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 4; ++nr)
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mr], bf[nr], acc[mr][nr], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // transpose through LDS, then mask + reduce + store row-contiguously
+  __syncthreads();
+  bf16* ct = reinterpret_cast<bf16*>(smem) + wave * wrows * 64;
+#pragma unroll
+  for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      const int col = nr * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int trow = mr * 16 + (lane >> 4) * 4 + r;
+        ct[trow * 64 + col] = f2bf(acc[mr][nr][r]);
+      }
+    }
+  __syncthreads();
+
+  // per-lane channel class: piece = i*64 + lane, 64 % 8 == 0 => this lane
+  // always touches channels c8 = (lane % 8) * 8 .. +8
+  const int c8 = n0 + wc * 64 + (lane % 8) * 8;
+  float sdb[8] = {}, sdg[8] = {};
+  float mn[8], rs[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mn[j] = mean[c8 + j];
+    rs[j] = rstd[c8 + j];
+  }
+  const int npieces = wrows * 64 / 8 / WAVE;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    if (i >= npieces) break;
+    const int piece = i * WAVE + lane;
+    const int trow = piece / 8;
+    const int tcol = (piece % 8) * 8;
+    const int64_t row = m0 + wr * wrows + trow;
+    const int64_t off = row * N + n0 + wc * 64 + tcol;
+    const bf16x8 gv = *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+    const bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + off);
+    const bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + off);
+    bf16x8 gp;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = (float)gv[j];
+      if ((float)yv[j] <= 0.f) g = 0.f;
+      gp[j] = (__bf16)g;
+      sdb[j] += g;
+      sdg[j] += g * ((float)zv[j] - mn[j]) * rs[j];
+    }
+    *reinterpret_cast<bf16x8*>(C + off) = gp;
+  }
+  // reduce across the 8 lanes of each channel class (xor 8/16/32 preserve
+  // lane % 8), then one atomic pair per channel from lanes 0..7
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    for (int m = 8; m <= 32; m <<= 1) {
+      sdb[j] += __shfl_xor(sdb[j], m, WAVE);
+      sdg[j] += __shfl_xor(sdg[j], m, WAVE);
+    }
+  }
+  if (lane < 8) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&dbeta[c8 + j], sdb[j]);
+      atomicAdd(&dgamma[c8 + j], sdg[j]);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int launch_gemm_nt_dgrad_epi(void* stream, const void* A,
+                                        const void* Bt, void* C, int64_t M,
+                                        int64_t N, int64_t K, const void* y,
+                                        const void* z, const float* mean,
+                                        const float* rstd, float* dbeta,
+                                        float* dgamma) {
+  if (M % 128 != 0 || K % 64 != 0 || N % 64 != 0)
+    return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int BNT = (N % 128 == 0) ? 128 : 64;
+  const int smem = (16384 + BNT * 128) * 2;
+  dim3 grid((unsigned)(M / 128), (unsigned)(N / BNT));
+  if (BNT == 128)
+    gemm_nt_dgrad_epi_kernel<128><<<grid, 256, smem, s>>>(
+        (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,
+        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma);
+  else
+    gemm_nt_dgrad_epi_kernel<64><<<grid, 256, smem, s>>>(
+        (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,
+        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma);
+  HIP_CHECK_LAST();
+  return 0;
+}

@@ -135,3 +135,12 @@ extern "C"
 int launch_conv1x1_strided(void* stream, const void* x, const void* w,
                            void* out, int64_t N, int64_t H, int64_t W,
                            int64_t Ci, int64_t Co, int64_t stride);
+#ifdef __cplusplus
+extern "C"
+#endif
+// dgrad GEMM + fused BN-backward epilogue (round-2 prototype): C = relu-
+// masked A@Bt^T, dbeta/dgamma partials accumulated (pre-zeroed, fp32).
+int launch_gemm_nt_dgrad_epi(void* stream, const void* A, const void* Bt,
+                             void* C, int64_t M, int64_t N, int64_t K,
+                             const void* y, const void* z, const float* mean,
+                             const float* rstd, float* dbeta, float* dgamma);

@@ -608,3 +608,30 @@ def test_two_crop_augment_uint8_gpu():
     out = _C.two_crop_augment(u8.cuda(), params.cuda(), S, False, False)
     ref = _augment_torch(u8.float() / 255.0, params, S)
     assert (out.cpu() - ref).abs().max() < 5e-2
+
+
+@experimental
+@pytest.mark.parametrize("M,N,K", [(1536, 128, 128), (256, 64, 64)])
+def test_gemm_nt_dgrad_epi(M, N, K):
+    """dgrad GEMM whose epilogue applies the consumer ReLU mask and
+    accumulates the BN-backward channel reductions."""
+    from moco_amd import _C
+    torch.manual_seed(23)
+    A = (torch.randn(M, K) * 0.3).cuda().to(torch.bfloat16)
+    Bt = (torch.randn(N, K) * 0.1).cuda().to(torch.bfloat16)
+    y = torch.randn(M, N).cuda().to(torch.bfloat16)
+    z = torch.randn(M, N).cuda().to(torch.bfloat16)
+    mean = torch.randn(N).cuda() * 0.1
+    rstd = (torch.rand(N) + 0.5).cuda()
+    C, dbeta, dgamma = _C.gemm_nt_dgrad_epi(A, Bt, y, z, mean, rstd)
+    dy = A.float() @ Bt.float().t()
+    mask = (y.float() > 0).float()
+    gp = (dy * mask).to(torch.bfloat16).float()
+    tol = 0.05 * max(1.0, K ** 0.5 * 0.03)
+    assert (C.float() - gp).abs().max() < tol
+    # reductions use the pre-rounding fp32 g' (dy*mask), bf16-rounded stores
+    ref_db = (dy * mask).sum(0)
+    zhat = (z.float() - mean[None, :]) * rstd[None, :]
+    ref_dg = (dy * mask * zhat).sum(0)
+    assert torch.allclose(dbeta.cpu(), ref_db.cpu(), rtol=5e-3, atol=0.5)
+    assert torch.allclose(dgamma.cpu(), ref_dg.cpu(), rtol=8e-3, atol=1.0)
