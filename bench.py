@@ -31,8 +31,8 @@ BASELINE_NODE_IMG_S = 1340.0  # 8x V100 derived
 def get_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-per-gpu", type=int, default=256)
     p.add_argument("--moco-k", type=int, default=65536)
     p.add_argument("--arch", default="resnet50")

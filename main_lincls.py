@@ -79,7 +79,10 @@ def train_transform_batch(raw: torch.Tensor, crop: int,
     reference lincls train aug).  Runs on whatever device `raw` lives on
     (the HIP augmentation kernels when on GPU)."""
     B, _, H, W = raw.shape
-    params = sample_two_crop_params(B, H, W, crop, aug_plus=False)
+    # torchvision-default crop scale (0.08, 1.0) — the reference lincls uses
+    # plain RandomResizedCrop, not MoCo's (0.2, 1.0) pretrain scale
+    params = sample_two_crop_params(B, H, W, crop, aug_plus=False,
+                                    scale=(0.08, 1.0))
     params = params[:B]
     # disable jitter/grayscale (lincls aug has none)
     params[:, 5] = 0.0   # gray

@@ -235,6 +235,11 @@ def main_worker(gpu, ngpus_per_node, args):
         else:
             train_epoch(loader, stepper, epoch, args, metrics=metrics)
 
+        if distributed:
+            # DDP runs with broadcast_buffers=False; average BN running
+            # stats across ranks so the checkpoint reflects the whole job
+            from moco_amd.parallel import sync_bn_running_stats
+            sync_bn_running_stats(model.module)
         if not distributed or dist.get_rank() == 0:
             raw_model = model.module if distributed else model
             # checkpoint contract: DDP-prefixed keys (module.encoder_q.* ...)

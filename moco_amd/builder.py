@@ -129,6 +129,11 @@ class MoCo(nn.Module):
         if step is None:
             step = self._step
             self._step += 1
+        # the shuffle permutation is derived from (seed, step) on every rank
+        # independently — verify the counters agree periodically (a desynced
+        # rank would otherwise corrupt the key pairing SILENTLY)
+        if step % 64 == 0:
+            parallel.assert_step_sync(step, im_q.device)
 
         with torch.no_grad():
             # start the key-batch exchange on the comm stream FIRST so the
@@ -153,8 +158,14 @@ class MoCo(nn.Module):
             # takes the hipBLASLt fallback inside contrastive_logits
             q = q.to(torch.bfloat16)
             k = k.to(torch.bfloat16)
+        # tell the op which queue block the enqueue below will overwrite so
+        # its backward snapshots only those columns (not the whole ring)
+        if self._ptr_host is None:
+            self._ptr_host = int(self.queue_ptr.item())
+        span = k.shape[0] * parallel.get_world_size()
         logits = contrastive_logits(q, k, self._working_queue(), self.queue,
-                                    self.T)
+                                    self.T, snap_ptr=self._ptr_host,
+                                    snap_span=span)
         labels = torch.zeros(logits.shape[0], dtype=torch.long,
                              device=logits.device)
         self._dequeue_and_enqueue(k)

@@ -41,11 +41,14 @@ GRAY_R, GRAY_G, GRAY_B = 0.299, 0.587, 0.114
 def sample_two_crop_params(batch: int, src_h: int, src_w: int, out_size: int,
                            aug_plus: bool,
                            generator: Optional[torch.Generator] = None,
-                           device: str = "cpu") -> torch.Tensor:
+                           device: str = "cpu",
+                           scale=(0.2, 1.0)) -> torch.Tensor:
     """Sample augmentation parameters for both crops of a batch.
 
     Returns [2*batch, 12] float32 (crop-major: rows [0:B] = crop q,
-    rows [B:2B] = crop k).
+    rows [B:2B] = crop k).  `scale` is the RandomResizedCrop area range:
+    (0.2, 1.0) for MoCo pretraining (reference main_moco.py:238,253),
+    (0.08, 1.0) = torchvision default for the lincls train transform.
     """
     n = 2 * batch
     g = generator
@@ -53,14 +56,15 @@ def sample_two_crop_params(batch: int, src_h: int, src_w: int, out_size: int,
     def rand(*shape):
         return torch.rand(*shape, generator=g)
 
-    # RandomResizedCrop(out, scale=(0.2, 1.0), ratio=(3/4, 4/3)) - 10 trials
+    # RandomResizedCrop(out, scale, ratio=(3/4, 4/3)) - 10 trials
     area = float(src_h * src_w)
     log_r = (math.log(3.0 / 4.0), math.log(4.0 / 3.0))
     w = torch.empty(n)
     h = torch.empty(n)
     ok = torch.zeros(n, dtype=torch.bool)
+    s_lo, s_hi = scale
     for _ in range(10):
-        target_area = area * (0.2 + 0.8 * rand(n))
+        target_area = area * (s_lo + (s_hi - s_lo) * rand(n))
         aspect = torch.exp(log_r[0] + (log_r[1] - log_r[0]) * rand(n))
         tw = torch.sqrt(target_area * aspect).round()
         th = torch.sqrt(target_area / aspect).round()

@@ -84,7 +84,8 @@ class _ContrastiveLogits(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q: torch.Tensor, k: torch.Tensor,
                 queue_kc: torch.Tensor, queue_ck: torch.Tensor,
-                T: float) -> torch.Tensor:
+                T: float, snap_ptr: Optional[int] = None,
+                snap_span: Optional[int] = None) -> torch.Tensor:
         q = q.contiguous()
         k = k.contiguous()
         # MFMA kernel constraints (non-standard dims run the hipBLASLt
@@ -98,21 +99,48 @@ class _ContrastiveLogits(torch.autograd.Function):
             logits = ext().contrastive_fwd(q, k, queue_kc.contiguous(),
                                            1.0 / T)
         else:
-            qf, kf, Qf = q.float(), k.float(), queue_kc.float()
+            # negatives from the canonical fp32 queue (queue_ck), NOT the
+            # bf16 working copy: an explicit fp32 run must keep fp32
+            # precision, and the backward below contracts against queue_ck
+            # — forward and backward must see the same negatives
+            qf, kf = q.float(), k.float()
             l_pos = (qf * kf).sum(dim=1, keepdim=True)
-            l_neg = qf @ Qf.t()
+            l_neg = qf @ queue_ck.float()
             logits = torch.cat([l_pos, l_neg], dim=1) / T
-        # snapshot: the enqueue that follows this forward overwrites B rows of
-        # the ring buffer in place (reference dodges this with
-        # queue.clone().detach() at builder.py:149)
-        ctx.save_for_backward(k, queue_ck.clone())
+        # snapshot: the enqueue that follows this forward overwrites
+        # `snap_span` columns of the ring buffer at `snap_ptr` in place
+        # (reference dodges this with queue.clone().detach() at
+        # builder.py:149, a full-queue clone every step).  When the caller
+        # tells us which block the enqueue will hit, save ONLY those columns
+        # (C x global-batch fp32, ~131 KB at the reference config, vs 33 MB
+        # for the clone at K=65536) and splice them back around the backward
+        # GEMM.  Without that info (direct op-level use), full clone.
+        if snap_ptr is not None and snap_span is not None:
+            cols = queue_ck[:, snap_ptr:snap_ptr + snap_span].clone()
+            ctx.save_for_backward(k, cols)
+            ctx.queue_ck_live = queue_ck  # plain attr: bypasses the version
+            # counter on purpose — we restore the mutated block ourselves
+            ctx.snap = (snap_ptr, snap_span)
+        else:
+            ctx.save_for_backward(k, queue_ck.clone())
+            ctx.snap = None
         ctx.inv_T = 1.0 / T
         ctx.q_dtype = q.dtype
         return logits
 
     @staticmethod
     def backward(ctx, grad_logits: torch.Tensor):
-        k, queue_ck = ctx.saved_tensors
+        k, saved = ctx.saved_tensors
+        if ctx.snap is not None:
+            ptr, span = ctx.snap
+            queue_ck = ctx.queue_ck_live
+            # splice the forward-time columns back in for the GEMM, then
+            # restore the live (post-enqueue) contents.  Safe: backward runs
+            # on the forward's stream before the next step's enqueue.
+            cur = queue_ck[:, ptr:ptr + span].clone()
+            queue_ck[:, ptr:ptr + span] = saved
+        else:
+            queue_ck = saved
         if use_hip(grad_logits) and ctx.shapes_ok:
             gq = ext().contrastive_bwd(grad_logits.contiguous(), k,
                                        queue_ck.contiguous(), ctx.inv_T)
@@ -120,13 +148,17 @@ class _ContrastiveLogits(torch.autograd.Function):
             g = grad_logits.float()
             gq = (g[:, :1] * k.float() +
                   g[:, 1:] @ queue_ck.float().t()) * ctx.inv_T
-        return gq.to(ctx.q_dtype), None, None, None, None
+        if ctx.snap is not None:
+            queue_ck[:, ptr:ptr + span] = cur
+        return gq.to(ctx.q_dtype), None, None, None, None, None, None
 
 
 def contrastive_logits(q: torch.Tensor, k: torch.Tensor,
                        queue_kc: torch.Tensor, queue_ck: torch.Tensor,
-                       T: float) -> torch.Tensor:
-    return _ContrastiveLogits.apply(q, k, queue_kc, queue_ck, T)
+                       T: float, snap_ptr: Optional[int] = None,
+                       snap_span: Optional[int] = None) -> torch.Tensor:
+    return _ContrastiveLogits.apply(q, k, queue_kc, queue_ck, T, snap_ptr,
+                                    snap_span)
 
 
 # ---------------------------------------------------------------------------

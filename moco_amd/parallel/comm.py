@@ -64,8 +64,43 @@ def _backend_supports_a2a() -> bool:
     return dist.get_backend() in ("nccl",)  # nccl == RCCL on ROCm
 
 
+class A2APlan:
+    """Host-side plan for one permutation exchange, computed once per step.
+
+    For the shuffle/unshuffle pair the permutation and its inverse are both
+    known up front (step_permutation), so neither direction needs the
+    O(total log total) argsorts of a from-scratch plan: `dest_perm` (global
+    src row -> global dest slot) and `src_of_slot` (its inverse) are passed
+    in, and only two length-n argsort/bincounts remain per direction.
+    """
+
+    def __init__(self, dest_perm: torch.Tensor, src_of_slot: torch.Tensor,
+                 world: int, n: int, rank: int):
+        total = world * n
+        assert dest_perm.numel() == total
+        dest_rank = torch.div(dest_perm, n, rounding_mode="floor")
+        my_dest_rank = dest_rank[rank * n:(rank + 1) * n]
+        self.send_order = torch.argsort(my_dest_rank, stable=True)
+        self.send_splits = torch.bincount(my_dest_rank,
+                                          minlength=world).tolist()
+        mine_src = src_of_slot[rank * n:(rank + 1) * n]
+        src_rank_of_mine = torch.div(mine_src, n, rounding_mode="floor")
+        self.recv_splits = torch.bincount(src_rank_of_mine,
+                                          minlength=world).tolist()
+        # recv rows arrive grouped by source rank; within each group in the
+        # sender's send order = ascending global SOURCE id (the stable sort
+        # by dest rank keeps rows in local = global source order). So
+        # arrival order = my slots sorted by (src_rank, src_global_id):
+        order_within = torch.argsort(src_rank_of_mine * total + mine_src)
+        pos_of_slot = torch.empty(n, dtype=torch.long)
+        pos_of_slot[order_within] = torch.arange(n)
+        self.pos_of_slot = pos_of_slot
+        self.n = n
+
+
 @torch.no_grad()
-def all_to_all_rows(x: torch.Tensor, dest_perm: torch.Tensor) -> torch.Tensor:
+def all_to_all_rows(x: torch.Tensor, dest_perm: torch.Tensor,
+                    plan: Optional[A2APlan] = None) -> torch.Tensor:
     """Exchange rows so that output row i on the destination layout equals
     global row dest_perm maps — generic permutation exchange.
 
@@ -82,8 +117,6 @@ def all_to_all_rows(x: torch.Tensor, dest_perm: torch.Tensor) -> torch.Tensor:
         return x[torch.argsort(dest_perm)] if dest_perm is not None else x
 
     rank = get_rank()
-    total = world * n
-    assert dest_perm.numel() == total
 
     if not _backend_supports_a2a():
         gathered = concat_all_gather(x)
@@ -93,66 +126,119 @@ def all_to_all_rows(x: torch.Tensor, dest_perm: torch.Tensor) -> torch.Tensor:
         mine = src_of_slot[rank * n:(rank + 1) * n]
         return gathered[mine].contiguous()
 
-    dest = dest_perm  # global src row g -> global dest slot
-    dest_rank = torch.div(dest, n, rounding_mode="floor")
-    # --- send side: my rows g in [rank*n, rank*n+n), ordered by dest rank
-    my_rows = torch.arange(rank * n, rank * n + n)
-    my_dest_rank = dest_rank[my_rows]
-    send_order = torch.argsort(my_dest_rank, stable=True)
-    send_splits = torch.bincount(my_dest_rank, minlength=world).tolist()
-    # --- recv side: rows destined for me, grouped by their source rank
-    src_of_slot = torch.argsort(dest)          # global slot -> global src row
-    mine_src = src_of_slot[rank * n:(rank + 1) * n]
-    src_rank_of_mine = torch.div(mine_src, n, rounding_mode="floor")
-    recv_splits = torch.bincount(src_rank_of_mine, minlength=world).tolist()
+    if plan is None:
+        plan = A2APlan(dest_perm, torch.argsort(dest_perm), world, n, rank)
 
-    send_buf = x[send_order.to(x.device)].contiguous()
+    send_buf = x[plan.send_order.to(x.device)].contiguous()
     recv_buf = torch.empty_like(send_buf)
     dist.all_to_all_single(recv_buf, send_buf,
-                           output_split_sizes=recv_splits,
-                           input_split_sizes=send_splits)
+                           output_split_sizes=plan.recv_splits,
+                           input_split_sizes=plan.send_splits)
+    return recv_buf[plan.pos_of_slot.to(x.device)].contiguous()
 
-    # recv_buf rows arrive grouped by source rank; within each source-rank
-    # group they are in the sender's send order = ascending global SOURCE id
-    # (the sender's stable sort by dest rank keeps its rows in local = global
-    # source order).  So arrival order = my slots sorted by
-    # (src_rank, src_global_id):
-    order_within = torch.argsort(src_rank_of_mine * total + mine_src)
-    # arrival position j carries the row for my slot order_within[j]
-    pos_of_slot = torch.empty(n, dtype=torch.long)
-    pos_of_slot[order_within] = torch.arange(n)
-    return recv_buf[pos_of_slot.to(x.device)].contiguous()
+
+class UnshuffleHandle:
+    """Opaque handle returned by shuffle_exchange: the unshuffle index plus
+    the precomputed reverse A2A plan (so the unshuffle pays no plan cost).
+    Index-compatible with the raw idx_unshuffle tensor for tests/tools."""
+
+    def __init__(self, idx_unshuffle: torch.Tensor,
+                 plan: Optional[A2APlan] = None):
+        self.idx_unshuffle = idx_unshuffle
+        self.plan = plan
+
+    def numel(self):
+        return self.idx_unshuffle.numel()
 
 
 @torch.no_grad()
-def shuffle_exchange(x: torch.Tensor, step: int, seed: int
-                     ) -> Tuple[torch.Tensor, torch.Tensor]:
+def shuffle_exchange(x: torch.Tensor, step: int, seed: int):
     """Batch-shuffle for shuffling-BN (reference _batch_shuffle_ddp,
-    builder.py:76-103).  Returns (shuffled local rows, idx_unshuffle)."""
+    builder.py:76-103).  Returns (shuffled local rows, unshuffle handle)."""
     world = get_world_size()
     n = x.shape[0]
     idx_shuffle, idx_unshuffle = step_permutation(step, world * n, seed,
                                                   x.device)
     if not is_distributed():
-        return x[idx_shuffle.to(x.device)], idx_unshuffle
-    # global source row g must land at global slot p where
-    # idx_shuffle[p] == g  =>  dest_perm = argsort? idx_shuffle maps slot->src,
-    # so dest of src g is the slot p with idx_shuffle[p] == g:
-    dest_perm = torch.argsort(idx_shuffle)
-    return all_to_all_rows(x, dest_perm), idx_unshuffle
+        return x[idx_shuffle.to(x.device)], UnshuffleHandle(idx_unshuffle)
+    # idx_shuffle maps slot->src, so dest slot of src g is argsort(idx_shuffle)
+    # = idx_unshuffle; the inverse map (src_of_slot) is idx_shuffle itself.
+    if _backend_supports_a2a():
+        rank = get_rank()
+        fwd = A2APlan(idx_unshuffle, idx_shuffle, world, n, rank)
+        rev = A2APlan(idx_shuffle, idx_unshuffle, world, n, rank)
+        out = all_to_all_rows(x, idx_unshuffle, plan=fwd)
+        return out, UnshuffleHandle(idx_unshuffle, plan=rev)
+    return (all_to_all_rows(x, idx_unshuffle),
+            UnshuffleHandle(idx_unshuffle))
 
 
 @torch.no_grad()
-def unshuffle_exchange(x: torch.Tensor,
-                       idx_unshuffle: torch.Tensor) -> torch.Tensor:
+def unshuffle_exchange(x: torch.Tensor, idx_unshuffle) -> torch.Tensor:
     """Inverse of shuffle_exchange (reference _batch_unshuffle_ddp,
-    builder.py:106-123)."""
-    world = get_world_size()
-    n = x.shape[0]
+    builder.py:106-123).  Accepts the UnshuffleHandle from shuffle_exchange
+    (reuses its precomputed plan) or a raw idx_unshuffle tensor."""
+    plan = None
+    if isinstance(idx_unshuffle, UnshuffleHandle):
+        plan = idx_unshuffle.plan
+        idx_unshuffle = idx_unshuffle.idx_unshuffle
     if not is_distributed():
         return x[idx_unshuffle.to(x.device)]
-    dest_perm = torch.argsort(idx_unshuffle)
-    return all_to_all_rows(x, dest_perm)
+    dest_perm = torch.argsort(idx_unshuffle) if plan is None else None
+    return all_to_all_rows(x, dest_perm, plan=plan)
+
+
+@torch.no_grad()
+def sync_bn_running_stats(module: torch.nn.Module) -> None:
+    """Average BatchNorm running_mean/running_var across ranks.
+
+    DDP is created with broadcast_buffers=False (the queue must stay
+    rank-deterministic, and a per-step buffer broadcast is redundant), so
+    per-rank BN running stats drift apart over an epoch; call this before
+    checkpointing/validation so the saved stats reflect the whole job.
+    One flat all-reduce for all BN buffers.
+    """
+    if not is_distributed():
+        return
+    bufs = []
+    for m in module.modules():
+        if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+            if m.running_mean is not None:
+                bufs.append(m.running_mean)
+            if m.running_var is not None:
+                bufs.append(m.running_var)
+    if not bufs:
+        return
+    flat = torch.cat([b.detach().reshape(-1) for b in bufs])
+    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    flat.div_(get_world_size())
+    off = 0
+    for b in bufs:
+        n = b.numel()
+        b.copy_(flat[off:off + n].view_as(b))
+        off += n
+
+
+@torch.no_grad()
+def assert_step_sync(step: int, device: torch.device) -> None:
+    """All-gather the step counter and assert every rank agrees.
+
+    Guards the shuffling-BN contract: the shuffle/unshuffle permutation is
+    derived from a shared (seed, step) pair, so a single rank skipping a
+    step would silently corrupt the key/positive pairing.  Called every
+    STEP_SYNC_EVERY steps from MoCo.forward (an 8-byte latency-bound
+    collective, amortized to nothing).
+    """
+    if not is_distributed():
+        return
+    t = torch.tensor([step], dtype=torch.long, device=device)
+    out = [torch.empty_like(t) for _ in range(get_world_size())]
+    dist.all_gather(out, t)
+    steps = [int(o.item()) for o in out]
+    if any(s != step for s in steps):
+        raise RuntimeError(
+            f"shuffling-BN step counters desynced across ranks: {steps} "
+            "(rank-local step drift corrupts the shuffle/unshuffle pairing)")
 
 
 # ---------------------------------------------------------------------------
