@@ -18,6 +18,7 @@ from typing import Callable, List, Optional, Type, Union
 import torch
 import torch.nn as nn
 
+from ..ops.block import fused_bottleneck
 from ..ops.bn import bn_act
 from ..ops.conv1x1 import conv1x1 as conv1x1_op
 from ..ops.pool import stem_maxpool
@@ -78,6 +79,11 @@ class Bottleneck(nn.Module):
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # hot path: the whole block as fused MFMA GEMMs with BN stats/apply
+        # folded into prologues/epilogues (ops/block.py)
+        out = fused_bottleneck(x, self)
+        if out is not None:
+            return out
         identity = x
         out = bn_act(conv1x1_op(x, self.conv1), self.bn1, relu=True)
         out = bn_act(self.conv2(out), self.bn2, relu=True)

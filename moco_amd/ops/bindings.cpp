@@ -263,25 +263,34 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor g, torch::Tensor y,
 }
 
 // ---------------------------------------------------------------------------
-// bf16 NT-GEMM (1x1-conv forward/dgrad): C[M,N] = A[M,K] @ Bt[N,K]^T
-torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt) {
+// bf16 NT-GEMM (1x1-conv forward/dgrad): C[M,N] = A[M,K] @ Bt[N,K]^T.
+// c_add (optional): epilogue C += c_add — fuses the residual-path gradient
+// accumulation into the dgrad GEMM store tail.
+torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt,
+                           c10::optional<torch::Tensor> c_add) {
   CHECK_CUDA_CONTIG(A);
   CHECK_CUDA_CONTIG(Bt);
   TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && A.size(1) == Bt.size(1));
   const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  if (c_add.has_value())
+    TORCH_CHECK(is_bf16(*c_add) && c_add->is_contiguous() &&
+                c_add->numel() == M * N);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
   // LDS-staged kernel for conforming shapes, naive fragment-load fallback
-  CHECK_LAUNCH(launch_gemm_nt_lds_bf16(cur_stream(), A.data_ptr(),
-                                       Bt.data_ptr(), C.data_ptr(), M, N,
-                                       K));
+  CHECK_LAUNCH(launch_gemm_nt_lds_bf16(
+      cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
+      c_add.has_value() ? c_add->data_ptr() : nullptr));
   return C;
 }
 
-// Round-2 prototype: GEMM with fused BN-apply (A path) / BN-stats (output).
+// GEMM with fused BN-apply (A path) / BN-stats (output) — the conv+BN hot
+// path.  want_aout materializes the transformed A operand (consumer conv
+// backward wgrad input); c_add fuses an epilogue add.
 std::vector<torch::Tensor> gemm_nt_fused(
     torch::Tensor A, torch::Tensor Bt, c10::optional<torch::Tensor> a_scale,
-    c10::optional<torch::Tensor> a_shift, bool a_relu, bool want_stats) {
+    c10::optional<torch::Tensor> a_shift, bool a_relu, bool want_stats,
+    bool want_aout, c10::optional<torch::Tensor> c_add) {
   CHECK_CUDA_CONTIG(A);
   CHECK_CUDA_CONTIG(Bt);
   TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && A.size(1) == Bt.size(1));
@@ -289,21 +298,102 @@ std::vector<torch::Tensor> gemm_nt_fused(
   TORCH_CHECK(a_scale.has_value() == a_shift.has_value());
   if (a_scale.has_value())
     TORCH_CHECK(a_scale->numel() == K && a_shift->numel() == K);
+  TORCH_CHECK(!want_aout || a_scale.has_value(),
+              "want_aout requires the BN-apply prologue");
+  if (c_add.has_value())
+    TORCH_CHECK(is_bf16(*c_add) && c_add->is_contiguous() &&
+                c_add->numel() == M * N);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
-  torch::Tensor ssum, ssq;
+  torch::Tensor ssum, ssq, aout;
   if (want_stats) {
     ssum = torch::zeros({N}, A.options().dtype(torch::kFloat32));
     ssq = torch::zeros({N}, A.options().dtype(torch::kFloat32));
   }
+  if (want_aout) aout = torch::empty({M, K}, A.options());
   CHECK_LAUNCH(launch_gemm_nt_fused_bf16(
       cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
       a_scale.has_value() ? a_scale->data_ptr<float>() : nullptr,
       a_shift.has_value() ? a_shift->data_ptr<float>() : nullptr, a_relu,
       want_stats ? ssum.data_ptr<float>() : nullptr,
-      want_stats ? ssq.data_ptr<float>() : nullptr));
-  if (want_stats) return {C, ssum, ssq};
-  return {C};
+      want_stats ? ssq.data_ptr<float>() : nullptr,
+      want_aout ? aout.data_ptr() : nullptr,
+      c_add.has_value() ? c_add->data_ptr() : nullptr));
+  std::vector<torch::Tensor> out = {C};
+  if (want_stats) {
+    out.push_back(ssum);
+    out.push_back(ssq);
+  }
+  if (want_aout) out.push_back(aout);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Granular BN entry points for the conv+BN fused block path.
+
+// stats only (1 read): mean/rstd + running-stats update
+std::vector<torch::Tensor> bn_stats(torch::Tensor x,
+                                    c10::optional<torch::Tensor> rmean,
+                                    c10::optional<torch::Tensor> rvar,
+                                    double eps, double momentum) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x));
+  const int64_t C = x.size(1);
+  const int64_t R = x.numel() / C;
+  at::cuda::CUDAGuard guard(x.device());
+  auto f32 = x.options().dtype(torch::kFloat32);
+  auto part_s = torch::empty({1024, C}, f32);
+  auto part_q = torch::empty({1024, C}, f32);
+  auto mean = torch::empty({C}, f32);
+  auto rstd = torch::empty({C}, f32);
+  CHECK_LAUNCH(launch_bn_stats(
+      cur_stream(), x.data_ptr(), part_s.data_ptr<float>(),
+      part_q.data_ptr<float>(), mean.data_ptr<float>(),
+      rstd.data_ptr<float>(),
+      rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
+      rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, (float)eps,
+      (float)momentum));
+  return {mean, rstd};
+}
+
+// finalize from the conv-GEMM epilogue's per-channel totals
+std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
+                                       int64_t R,
+                                       c10::optional<torch::Tensor> rmean,
+                                       c10::optional<torch::Tensor> rvar,
+                                       double eps, double momentum) {
+  CHECK_CUDA_CONTIG(ssum);
+  CHECK_CUDA_CONTIG(ssq);
+  const int64_t C = ssum.numel();
+  at::cuda::CUDAGuard guard(ssum.device());
+  auto mean = torch::empty({C}, ssum.options());
+  auto rstd = torch::empty({C}, ssum.options());
+  CHECK_LAUNCH(launch_bn_finalize(
+      cur_stream(), ssum.data_ptr<float>(), ssq.data_ptr<float>(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
+      rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, (float)eps,
+      (float)momentum));
+  return {mean, rstd};
+}
+
+// dx-only backward: upstream grad already relu-masked (by the dgrad-GEMM
+// epilogue), dbeta/dgamma already reduced there — one elementwise pass.
+torch::Tensor bn_bwd_dx_only(torch::Tensor gp, torch::Tensor z,
+                             torch::Tensor mean, torch::Tensor rstd,
+                             torch::Tensor gamma, torch::Tensor dbeta,
+                             torch::Tensor dgamma) {
+  TORCH_CHECK(gp.is_cuda() && is_bf16(gp) && is_bf16(z));
+  const int64_t C = z.size(1);
+  const int64_t R = z.numel() / C;
+  at::cuda::CUDAGuard guard(z.device());
+  auto dx = torch::empty_like(gp);
+  CHECK_LAUNCH(launch_bn_bwd_dx(
+      cur_stream(), gp.data_ptr(), gp.data_ptr(), z.data_ptr(),
+      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+      gamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+      dgamma.data_ptr<float>(), dx.data_ptr(), nullptr, R, C,
+      /*relu=*/false, /*train=*/true));
+  return dx;
 }
 
 // Round-2 prototype: 3x3 pad-1 conv (stride 1/2) as implicit MFMA GEMM,
@@ -464,9 +554,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_fwd_train", &bn_fwd_train);
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
+  mod.def("bn_stats", &bn_stats);
+  mod.def("bn_finalize", &bn_finalize);
+  mod.def("bn_bwd_dx_only", &bn_bwd_dx_only);
   mod.def("maxpool_fwd", &maxpool_fwd);
-  mod.def("gemm_nt_bf16", &gemm_nt_bf16);
-  mod.def("gemm_nt_fused", &gemm_nt_fused);
+  mod.def("gemm_nt_bf16", &gemm_nt_bf16, py::arg("A"), py::arg("Bt"),
+          py::arg("c_add") = c10::nullopt);
+  mod.def("gemm_nt_fused", &gemm_nt_fused, py::arg("A"), py::arg("Bt"),
+          py::arg("a_scale"), py::arg("a_shift"), py::arg("a_relu"),
+          py::arg("want_stats"), py::arg("want_aout") = false,
+          py::arg("c_add") = c10::nullopt);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv1x1_strided", &conv1x1_strided);
   mod.def("gemm_nt_dgrad_epi", &gemm_nt_dgrad_epi);

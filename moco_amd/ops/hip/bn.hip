@@ -128,6 +128,28 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
   }
 }
 
+// ---- finalize from channel TOTALS (conv-GEMM stats epilogue path) ---------
+// The fused conv GEMM accumulates per-output-channel (sum, sumsq) of its
+// fp32 accumulators directly into a [C] pair; this turns them into
+// mean/rstd and updates the running stats — one tiny launch per conv.
+__global__ __launch_bounds__(256) void bn_finalize_kernel(
+    const float* __restrict__ ssum, const float* __restrict__ ssq,
+    float* __restrict__ mean, float* __restrict__ rstd,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    int64_t R, int C, float eps, float momentum) {
+  const int c = blockIdx.x * 256 + threadIdx.x;
+  if (c >= C) return;
+  const float m = ssum[c] / (float)R;
+  const float var = fmaxf(ssq[c] / (float)R - m * m, 0.f);
+  mean[c] = m;
+  rstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+    const float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
 // ---- forward apply --------------------------------------------------------
 template <bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_apply_kernel(
@@ -306,6 +328,19 @@ extern "C" int launch_bn_stats(void* stream, const void* x, float* part_s,
   bn_stats_final_kernel<<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
       part_s, part_q, mean, rstd, running_mean, running_var, R, (int)C,
       P, eps, momentum);
+  HIP_CHECK_LAST();
+  return 0;
+}
+
+extern "C" int launch_bn_finalize(void* stream, const float* ssum,
+                                  const float* ssq, float* mean, float* rstd,
+                                  float* running_mean, float* running_var,
+                                  int64_t R, int64_t C, float eps,
+                                  float momentum) {
+  hipStream_t s = (hipStream_t)stream;
+  bn_finalize_kernel<<<dim3((unsigned)((C + 255) / 256)), 256, 0, s>>>(
+      ssum, ssq, mean, rstd, running_mean, running_var, R, (int)C, eps,
+      momentum);
   HIP_CHECK_LAST();
   return 0;
 }

@@ -121,20 +121,26 @@ __device__ __forceinline__ bf16x8 lds_frag(const char* smem, int row,
 // BNT = 128: 2x2 wave grid, 64x64 per wave (MR=NR=4)
 // BNT =  64: 4x1 wave grid, 32x64 per wave (MR=2, NR=4)
 // DBUF: double-buffered staging (off for single-K-tile problems)
-// APPLY (round-2 conv+BN-fusion prototype): transform every A element with
+// APPLY (conv+BN fusion): transform every A element with
 //   x' = relu?(x * scale[k] + shift[k])  — i.e. the BatchNorm apply of the
 //   PREVIOUS layer folded into this GEMM's fragment path (scale = rstd*gamma,
 //   shift = beta - mean*rstd*gamma, precomputed per channel on the host).
 // STATS: accumulate per-output-channel (sum, sumsq) of the fp32 accumulator
 //   tile into stats_sum/stats_sumsq (pre-zeroed, fp32 atomics) — the BN
 //   statistics of THIS layer's output for free.
+// a_out (runtime, with APPLY): materialize the transformed A operand (the
+//   consumer conv's backward wgrad needs it); only the blockIdx.y==0 / wc==0
+//   waves write, so each element is written exactly once.
+// c_add (runtime): epilogue C += c_add — fuses the residual-path gradient
+//   accumulation into the dgrad GEMM's store tail.
 template <int BNT, bool DBUF, bool APPLY, bool STATS>
 __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     bf16* __restrict__ C, int64_t M, int N, int K,
     const float* __restrict__ a_scale, const float* __restrict__ a_shift,
     bool a_relu, float* __restrict__ stats_sum,
-    float* __restrict__ stats_sumsq) {
+    float* __restrict__ stats_sumsq, bf16* __restrict__ a_out,
+    const bf16* __restrict__ c_add) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int MR = BNT == 128 ? 4 : 2;
   constexpr int ABYTES = BM * BK * 2;    // 16 KiB
@@ -189,6 +195,15 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
             if (a_relu) v = fmaxf(v, 0.f);
             af[mr][j] = (__bf16)v;
           }
+        if (a_out != nullptr && blockIdx.y == 0 && wc == 0) {
+          // each transformed A element written once: rows wr*wrows + mr*16 +
+          // (lane&15), cols c0..c0+8 (this lane's fragment span)
+#pragma unroll
+          for (int mr = 0; mr < MR; ++mr) {
+            const int64_t row = m0 + wr * wrows + mr * 16 + (lane & 15);
+            *reinterpret_cast<bf16x8*>(a_out + row * K + c0) = af[mr];
+          }
+        }
       }
 #pragma unroll
       for (int nr = 0; nr < 4; ++nr)
@@ -258,8 +273,14 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
     const int trow = piece / 8;
     const int tcol = (piece % 8) * 8;
     const int64_t row = m0 + wr * wrows + trow;
-    *reinterpret_cast<bf16x8*>(C + row * N + n0 + wc * 64 + tcol) =
-        *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+    const int64_t off = row * N + n0 + wc * 64 + tcol;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+    if (c_add != nullptr) {
+      const bf16x8 ad = *reinterpret_cast<const bf16x8*>(c_add + off);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = (__bf16)((float)v[j] + (float)ad[j]);
+    }
+    *reinterpret_cast<bf16x8*>(C + off) = v;
   }
 }
 
@@ -267,11 +288,14 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
 
 extern "C" int launch_gemm_nt_lds_bf16(void* stream, const void* A,
                                        const void* Bt, void* C, int64_t M,
-                                       int64_t N, int64_t K) {
+                                       int64_t N, int64_t K,
+                                       const void* c_add) {
   const bool n128 = (N % 128 == 0);
   const bool n64 = (N % 64 == 0);
-  if (M % 128 != 0 || K % 64 != 0 || !(n128 || n64))
+  if (M % 128 != 0 || K % 64 != 0 || !(n128 || n64)) {
+    if (c_add != nullptr) return (int)hipErrorInvalidValue;
     return launch_gemm_nt_bf16(stream, A, Bt, C, M, N, K);
+  }
   hipStream_t s = (hipStream_t)stream;
   const int BNT = n128 ? 128 : 64;
   const bool dbuf = (K / 64) > 1;
@@ -280,7 +304,8 @@ extern "C" int launch_gemm_nt_lds_bf16(void* stream, const void* A,
 #define GO(BNTv, DB)                                                       \
   gemm_nt_lds_kernel<BNTv, DB, false, false><<<grid, 256, smem, s>>>(      \
       (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,        \
-      nullptr, nullptr, false, nullptr, nullptr)
+      nullptr, nullptr, false, nullptr, nullptr, nullptr,                  \
+      (const bf16*)c_add)
   if (BNT == 128 && dbuf) GO(128, true);
   else if (BNT == 128) GO(128, false);
   else if (dbuf) GO(64, true);
@@ -299,9 +324,11 @@ extern "C" int launch_gemm_nt_fused_bf16(void* stream, const void* A,
                                          const float* a_scale,
                                          const float* a_shift, bool a_relu,
                                          float* stats_sum,
-                                         float* stats_sumsq) {
+                                         float* stats_sumsq, void* a_out,
+                                         const void* c_add) {
   if (M % 128 != 0 || K % 64 != 0 || (N % 64 != 0))
     return (int)hipErrorInvalidValue;
+  if (a_out != nullptr && a_scale == nullptr) return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
   const int BNT = (N % 128 == 0) ? 128 : 64;
   const bool dbuf = (K / 64) > 1;
@@ -312,7 +339,8 @@ extern "C" int launch_gemm_nt_fused_bf16(void* stream, const void* A,
 #define GO(BNTv, DB, AP, ST)                                               \
   gemm_nt_lds_kernel<BNTv, DB, AP, ST><<<grid, 256, smem, s>>>(            \
       (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,        \
-      a_scale, a_shift, a_relu, stats_sum, stats_sumsq)
+      a_scale, a_shift, a_relu, stats_sum, stats_sumsq, (bf16*)a_out,      \
+      (const bf16*)c_add)
 #define GO2(BNTv, DB)                                                      \
   do {                                                                     \
     if (apply && stats) GO(BNTv, DB, true, true);                          \

@@ -70,6 +70,12 @@ int launch_bn_bwd_dx(void* stream, const void* g, const void* y,
                      const float* gamma, const float* dbeta,
                      const float* dgamma, void* dx, void* dres, int64_t R,
                      int64_t C, bool relu, bool train);
+// finalize mean/rstd (+ running stats) from per-channel TOTAL (sum, sumsq)
+// as produced by the conv-GEMM stats epilogue
+int launch_bn_finalize(void* stream, const float* ssum, const float* ssq,
+                       float* mean, float* rstd, float* running_mean,
+                       float* running_var, int64_t R, int64_t C, float eps,
+                       float momentum);
 
 // ---- augment.hip ----------------------------------------------------------
 // src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,
@@ -108,16 +114,20 @@ int launch_gemm_nt_bf16(void* stream, const void* A, const void* Bt, void* C,
 #ifdef __cplusplus
 extern "C"
 #endif
+// c_add (nullable): epilogue C += c_add (requires the LDS-conforming shape)
 int launch_gemm_nt_lds_bf16(void* stream, const void* A, const void* Bt,
-                            void* C, int64_t M, int64_t N, int64_t K);
+                            void* C, int64_t M, int64_t N, int64_t K,
+                            const void* c_add);
 #ifdef __cplusplus
 extern "C"
 #endif
+// a_out (nullable, needs a_scale): materialized transformed A operand [M,K]
 int launch_gemm_nt_fused_bf16(void* stream, const void* A, const void* Bt,
                               void* C, int64_t M, int64_t N, int64_t K,
                               const float* a_scale, const float* a_shift,
                               bool a_relu, float* stats_sum,
-                              float* stats_sumsq);
+                              float* stats_sumsq, void* a_out,
+                              const void* c_add);
 #ifdef __cplusplus
 extern "C"
 #endif

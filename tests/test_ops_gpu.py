@@ -466,17 +466,14 @@ def test_single_crop_augment_vs_oracle():
 
 
 # ---------------------------------------------------------------------------
-# round-2 prototype: fused GEMM + BN apply/stats (env-gated: experimental)
+# fused GEMM + BN apply/stats (the conv+BN hot path) and remaining
+# round-2 prototypes.  All in the default GPU suite.
 # ---------------------------------------------------------------------------
 
-import os as _os
-
-experimental = pytest.mark.skipif(
-    _os.environ.get("MOCO_AMD_EXPERIMENTAL") != "1",
-    reason="round-2 prototype; enable with MOCO_AMD_EXPERIMENTAL=1")
+def experimental(f):  # graduated round-2 kernels: no longer env-gated
+    return f
 
 
-@experimental
 @pytest.mark.parametrize("M,N,K", [(1536, 128, 64), (128, 64, 256)])
 def test_gemm_nt_fused_apply_and_stats(M, N, K):
     from moco_amd import _C
@@ -498,7 +495,6 @@ def test_gemm_nt_fused_apply_and_stats(M, N, K):
                           atol=1.0)
 
 
-@experimental
 def test_gemm_nt_fused_no_options_matches_plain():
     from moco_amd import _C
     torch.manual_seed(17)
@@ -509,7 +505,6 @@ def test_gemm_nt_fused_no_options_matches_plain():
     assert torch.equal(C, ref)
 
 
-@experimental
 @pytest.mark.parametrize("N,Ci,H,Co,stride", [
     (8, 64, 16, 64, 1),     # M = 2048
     (8, 64, 16, 128, 1),
@@ -532,7 +527,6 @@ def test_conv3x3_fwd_vs_conv2d(N, Ci, H, Co, stride):
         (out.float() - ref).abs().max()
 
 
-@experimental
 def test_conv3x3_fused_apply_stats():
     """conv3x3 with the previous BN's apply folded in (neutral-page padding)
     + output stats."""
@@ -562,7 +556,6 @@ def test_conv3x3_fused_apply_stats():
                           atol=1.0)
 
 
-@experimental
 def test_conv3x3_dgrad_s1_matches_autograd():
     from moco_amd.ops.conv3x3 import conv3x3_dgrad_s1
     torch.manual_seed(20)
@@ -580,7 +573,6 @@ def test_conv3x3_dgrad_s1_matches_autograd():
         (dx.float() - x.grad).abs().max()
 
 
-@experimental
 @pytest.mark.parametrize("N,Ci,H,Co,stride", [
     (8, 256, 16, 512, 2), (2, 64, 32, 128, 2), (8, 64, 16, 64, 1)])
 def test_conv1x1_strided_vs_conv2d(N, Ci, H, Co, stride):
@@ -597,7 +589,6 @@ def test_conv1x1_strided_vs_conv2d(N, Ci, H, Co, stride):
                                                         Ci ** 0.5 * 0.025)
 
 
-@experimental
 def test_two_crop_augment_uint8_gpu():
     from moco_amd import _C
     from moco_amd.ops.augment import _augment_torch, sample_two_crop_params
@@ -610,7 +601,6 @@ def test_two_crop_augment_uint8_gpu():
     assert (out.cpu() - ref).abs().max() < 5e-2
 
 
-@experimental
 @pytest.mark.parametrize("M,N,K", [(1536, 128, 128), (256, 64, 64)])
 def test_gemm_nt_dgrad_epi(M, N, K):
     """dgrad GEMM whose epilogue applies the consumer ReLU mask and
@@ -635,3 +625,90 @@ def test_gemm_nt_dgrad_epi(M, N, K):
     ref_dg = (dy * mask * zhat).sum(0)
     assert torch.allclose(dbeta.cpu(), ref_db.cpu(), rtol=5e-3, atol=0.5)
     assert torch.allclose(dgamma.cpu(), ref_dg.cpu(), rtol=8e-3, atol=1.0)
+
+
+# ---------------------------------------------------------------------------
+# new round-2 kernel features: a_out / c_add epilogues, granular BN entries
+# ---------------------------------------------------------------------------
+
+def test_gemm_nt_fused_aout_materializes_transformed_a():
+    from moco_amd import _C
+    torch.manual_seed(24)
+    M, N, K = 1536, 128, 64
+    A = (torch.randn(M, K) * 0.7).cuda().to(torch.bfloat16)
+    Bt = (torch.randn(N, K) * 0.08).cuda().to(torch.bfloat16)
+    scale = (torch.rand(K) * 1.5 + 0.25).cuda()
+    shift = (torch.randn(K) * 0.3).cuda()
+    C, ssum, ssq, aout = _C.gemm_nt_fused(A, Bt, scale, shift, True, True,
+                                          True, None)
+    ref_a = torch.relu(A.float() * scale[None, :] + shift[None, :]) \
+        .to(torch.bfloat16)
+    assert torch.equal(aout, ref_a)
+    ref = ref_a.float() @ Bt.float().t()
+    assert (C.float() - ref).abs().max() < 0.08
+
+
+def test_gemm_nt_bf16_c_add_epilogue():
+    from moco_amd import _C
+    torch.manual_seed(25)
+    M, N, K = 256, 64, 128
+    A = (torch.randn(M, K) * 0.3).cuda().to(torch.bfloat16)
+    Bt = (torch.randn(N, K) * 0.1).cuda().to(torch.bfloat16)
+    add = torch.randn(M, N).cuda().to(torch.bfloat16)
+    C = _C.gemm_nt_bf16(A, Bt, add)
+    base = _C.gemm_nt_bf16(A, Bt)
+    ref = (base.float() + add.float()).to(torch.bfloat16)
+    assert torch.equal(C, ref)
+
+
+def test_bn_finalize_matches_stats():
+    from moco_amd import _C
+    torch.manual_seed(26)
+    R, C = 4096, 128
+    x = torch.randn(R, C).cuda().to(torch.bfloat16)
+    xf = x.float()
+    ssum = xf.sum(0)
+    ssq = (xf * xf).sum(0)
+    rm = torch.zeros(C).cuda()
+    rv = torch.ones(C).cuda()
+    mean, rstd = _C.bn_finalize(ssum, ssq, R, rm, rv, 1e-5, 0.1)
+    ref_mean = xf.mean(0)
+    ref_var = xf.var(0, unbiased=False)
+    assert torch.allclose(mean, ref_mean, atol=1e-3)
+    assert torch.allclose(rstd, torch.rsqrt(ref_var + 1e-5), rtol=1e-3)
+    assert torch.allclose(rm, 0.1 * ref_mean, atol=1e-3)
+    assert torch.allclose(rv, 0.9 + 0.1 * xf.var(0, unbiased=True),
+                          rtol=1e-3)
+
+
+def test_bn_stats_only_matches_batchnorm():
+    from moco_amd import _C
+    torch.manual_seed(27)
+    R, C = 2048, 64
+    x = torch.randn(R, C).cuda().to(torch.bfloat16)
+    rm = torch.zeros(C).cuda()
+    rv = torch.ones(C).cuda()
+    mean, rstd = _C.bn_stats(x, rm, rv, 1e-5, 0.1)
+    xf = x.float()
+    assert torch.allclose(mean, xf.mean(0), atol=2e-3)
+    assert torch.allclose(rstd, torch.rsqrt(xf.var(0, unbiased=False) + 1e-5),
+                          rtol=2e-3)
+    assert torch.allclose(rm, 0.1 * xf.mean(0), atol=2e-3)
+
+
+def test_bn_bwd_dx_only_matches_formula():
+    from moco_amd import _C
+    torch.manual_seed(28)
+    R, C = 1024, 64
+    gp = (torch.randn(R, C) * 0.1).cuda().to(torch.bfloat16)
+    z = torch.randn(R, C).cuda().to(torch.bfloat16)
+    mean = torch.randn(C).cuda() * 0.1
+    rstd = (torch.rand(C) + 0.5).cuda()
+    gamma = (torch.rand(C) + 0.5).cuda()
+    dbeta = torch.randn(C).cuda()
+    dgamma = torch.randn(C).cuda()
+    dx = _C.bn_bwd_dx_only(gp, z, mean, rstd, gamma, dbeta, dgamma)
+    zhat = (z.float() - mean[None, :]) * rstd[None, :]
+    ref = gamma[None, :] * rstd[None, :] * (
+        gp.float() - dbeta[None, :] / R - zhat * dgamma[None, :] / R)
+    assert (dx.float() - ref).abs().max() < 2e-2
