@@ -521,7 +521,8 @@ torch::Tensor two_crop_augment(torch::Tensor src, torch::Tensor params,
   const int64_t B = src.size(0), H = src.size(2), W = src.size(3);
   const int64_t M = params.size(0);
   at::cuda::CUDAGuard guard(src.device());
-  auto f32 = src.options();
+  // workspaces/out are fp32/bf16 regardless of the (possibly uint8) source
+  auto f32 = src.options().dtype(torch::kFloat32);
   auto out_opts = out_bf16 ? f32.dtype(torch::kBFloat16) : f32;
   torch::Tensor out = torch::empty({M, 3, S, S}, out_opts);
   if (channels_last)

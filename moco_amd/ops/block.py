@@ -61,6 +61,15 @@ def _bf16_weight(w: torch.Tensor) -> torch.Tensor:
         .contiguous()
 
 
+def _as_weight_grad(dw2d: torch.Tensor, strides) -> torch.Tensor:
+    """[Co, Ci] grad -> [Co, Ci, 1, 1] with the master conv weight's strides
+    (NCHW-contiguous and channels_last have identical storage order for 1x1
+    kernels, so this is a stride-only view — FusedSGD/DDP require grad
+    strides == param strides)."""
+    co, ci = dw2d.shape
+    return dw2d.as_strided((co, ci, 1, 1), tuple(strides))
+
+
 class _FusedBottleneck(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, identity, w1, gamma1, beta1, w2, gamma2, beta2,
@@ -116,7 +125,7 @@ class _FusedBottleneck(torch.autograd.Function):
                               mean1, rstd1, mean2, rstd2, mean3, rstd3,
                               scale2, shift2)
         ctx.meta = (N, H, W, Ho, Wo, stride, groups,
-                    identity is not None)
+                    identity is not None, w1.stride(), w3.stride())
         return _as4d(y, N, Ho, Wo)
 
     @staticmethod
@@ -124,7 +133,8 @@ class _FusedBottleneck(torch.autograd.Function):
         (x, z1, a1, z2, a2, z3, y, w1b, w2b, w3b, gamma1, gamma2, gamma3,
          mean1, rstd1, mean2, rstd2, mean3, rstd3, scale2,
          shift2) = ctx.saved_tensors
-        N, H, W, Ho, Wo, stride, groups, has_sep_identity = ctx.meta
+        (N, H, W, Ho, Wo, stride, groups, has_sep_identity, w1_strides,
+         w3_strides) = ctx.meta
         _C = ext()
 
         gy2d = _as2d(gy.contiguous(memory_format=torch.channels_last))
@@ -141,8 +151,7 @@ class _FusedBottleneck(torch.autograd.Function):
         dz2 = _C.bn_bwd_dx_only(gp2, z2, mean2, rstd2, gamma2, dbeta2,
                                 dgamma2)
         # conv3 wgrad (hipBLASLt TN over the materialized a2)
-        dw3 = torch.matmul(dz3.t(), a2).float() \
-            .view(w3b.shape[0], w3b.shape[1], 1, 1)
+        dw3 = _as_weight_grad(torch.matmul(dz3.t(), a2).float(), w3_strides)
 
         # conv2 backward (MIOpen)
         dz2_4d = _as4d(dz2, N, Ho, Wo)
@@ -162,8 +171,7 @@ class _FusedBottleneck(torch.autograd.Function):
         x2d = _as2d(x)
         dx2d = _C.gemm_nt_bf16(dz1, w1t,
                                None if has_sep_identity else dres)
-        dw1 = torch.matmul(dz1.t(), x2d).float() \
-            .view(w1b.shape[0], w1b.shape[1], 1, 1)
+        dw1 = _as_weight_grad(torch.matmul(dz1.t(), x2d).float(), w1_strides)
 
         dx = _as4d(dx2d, N, H, W)
         dident = _as4d(dres, N, Ho, Wo) if has_sep_identity else None
