@@ -307,8 +307,10 @@ std::vector<torch::Tensor> gemm_nt_fused(
   auto C = torch::empty({M, N}, A.options());
   torch::Tensor ssum, ssq, aout;
   if (want_stats) {
-    ssum = torch::zeros({N}, A.options().dtype(torch::kFloat32));
-    ssq = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+    // slot-striped [256, N] partials (see conv1x1.hip STAT_SLOTS);
+    // bn_finalize reduces the slots
+    ssum = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+    ssq = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
   }
   if (want_aout) aout = torch::empty({M, K}, A.options());
   CHECK_LAUNCH(launch_gemm_nt_fused_bf16(
@@ -355,7 +357,8 @@ std::vector<torch::Tensor> bn_stats(torch::Tensor x,
   return {mean, rstd};
 }
 
-// finalize from the conv-GEMM epilogue's per-channel totals
+// finalize from the conv-GEMM epilogue's slot-striped channel partials
+// ([P, C]; 1-D input means P = 1)
 std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
                                        int64_t R,
                                        c10::optional<torch::Tensor> rmean,
@@ -363,7 +366,8 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
                                        double eps, double momentum) {
   CHECK_CUDA_CONTIG(ssum);
   CHECK_CUDA_CONTIG(ssq);
-  const int64_t C = ssum.numel();
+  const int64_t C = ssum.dim() == 2 ? ssum.size(1) : ssum.numel();
+  const int64_t P = ssum.dim() == 2 ? ssum.size(0) : 1;
   at::cuda::CUDAGuard guard(ssum.device());
   auto mean = torch::empty({C}, ssum.options());
   auto rstd = torch::empty({C}, ssum.options());
@@ -371,8 +375,8 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
       cur_stream(), ssum.data_ptr<float>(), ssq.data_ptr<float>(),
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
-      rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, (float)eps,
-      (float)momentum));
+      rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, P,
+      (float)eps, (float)momentum));
   return {mean, rstd};
 }
 
@@ -470,14 +474,15 @@ std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
   TORCH_CHECK(mean.numel() == N && rstd.numel() == N);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
-  auto dbeta = torch::zeros({N}, A.options().dtype(torch::kFloat32));
-  auto dgamma = torch::zeros({N}, A.options().dtype(torch::kFloat32));
+  // slot-striped partials (conv1x1.hip STAT_SLOTS), reduced here
+  auto dbeta = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+  auto dgamma = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
   CHECK_LAUNCH(launch_gemm_nt_dgrad_epi(
       cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
       y.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), dbeta.data_ptr<float>(),
       dgamma.data_ptr<float>()));
-  return {C, dbeta, dgamma};
+  return {C, dbeta.sum(0), dgamma.sum(0)};
 }
 
 // ---------------------------------------------------------------------------

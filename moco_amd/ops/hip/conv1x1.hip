@@ -85,6 +85,7 @@ extern "C" int launch_gemm_nt_bf16(void* stream, const void* A,
 namespace {
 
 constexpr int BM = 128, BK = 64;
+constexpr int STAT_SLOTS = 256;  // slot-striped channel reductions
 // tile images: A [128 rows][64 k] bf16 = 16 KiB; B [BNT rows][64 k]
 
 __device__ __forceinline__ unsigned swz(unsigned row, unsigned kbyte) {
@@ -224,7 +225,12 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
   if (STATS) {
     // per-output-channel partial (sum, sumsq) of the fp32 accumulators.
     // Column c is held by lanes {l, l+16, l+32, l+48}: butterfly those
-    // four, then one atomic pair per (lane<16, nr).
+    // four, then one atomic pair per (lane<16, nr).  The destination is
+    // SLOT-STRIPED [STAT_SLOTS, N] (slot = blockIdx.x % STAT_SLOTS): with a
+    // flat [N] buffer every M-block of a big grid serializes on the same
+    // few cache lines (measured 4.8 ms on an M=802816, N=64 dispatch);
+    // striping cuts per-line contention ~STAT_SLOTS-fold and bn_finalize
+    // reduces the slots.
 #pragma unroll
     for (int nr = 0; nr < 4; ++nr) {
       float ssum = 0.f, ssq = 0.f;
@@ -241,9 +247,10 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_lds_kernel(
       ssum += __shfl_xor(ssum, 32, WAVE);
       ssq += __shfl_xor(ssq, 32, WAVE);
       if ((lane >> 4) == 0) {
+        const int64_t slot = (int64_t)(blockIdx.x % STAT_SLOTS) * N;
         const int c = n0 + wc * 64 + nr * 16 + lane;
-        atomicAdd(&stats_sum[c], ssum);
-        atomicAdd(&stats_sumsq[c], ssq);
+        atomicAdd(&stats_sum[slot + c], ssum);
+        atomicAdd(&stats_sumsq[slot + c], ssq);
       }
     }
   }
@@ -482,7 +489,8 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_dgrad_epi_kernel(
     *reinterpret_cast<bf16x8*>(C + off) = gp;
   }
   // reduce across the 8 lanes of each channel class (xor 8/16/32 preserve
-  // lane % 8), then one atomic pair per channel from lanes 0..7
+  // lane % 8), then one atomic pair per channel from lanes 0..7 into the
+  // slot-striped [STAT_SLOTS, N] partials (see the STATS epilogue note)
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     for (int m = 8; m <= 32; m <<= 1) {
@@ -491,10 +499,11 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_dgrad_epi_kernel(
     }
   }
   if (lane < 8) {
+    const int64_t slot = (int64_t)(blockIdx.x % STAT_SLOTS) * N;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&dbeta[c8 + j], sdb[j]);
-      atomicAdd(&dgamma[c8 + j], sdg[j]);
+      atomicAdd(&dbeta[slot + c8 + j], sdb[j]);
+      atomicAdd(&dgamma[slot + c8 + j], sdg[j]);
     }
   }
 }

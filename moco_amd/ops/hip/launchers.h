@@ -70,12 +70,12 @@ int launch_bn_bwd_dx(void* stream, const void* g, const void* y,
                      const float* gamma, const float* dbeta,
                      const float* dgamma, void* dx, void* dres, int64_t R,
                      int64_t C, bool relu, bool train);
-// finalize mean/rstd (+ running stats) from per-channel TOTAL (sum, sumsq)
-// as produced by the conv-GEMM stats epilogue
+// finalize mean/rstd (+ running stats) from the [P, C] slot-striped
+// (sum, sumsq) partials produced by the conv-GEMM stats epilogue
 int launch_bn_finalize(void* stream, const float* ssum, const float* ssq,
                        float* mean, float* rstd, float* running_mean,
-                       float* running_var, int64_t R, int64_t C, float eps,
-                       float momentum);
+                       float* running_var, int64_t R, int64_t C, int64_t P,
+                       float eps, float momentum);
 
 // ---- augment.hip ----------------------------------------------------------
 // src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,

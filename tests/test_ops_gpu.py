@@ -483,6 +483,7 @@ def test_gemm_nt_fused_apply_and_stats(M, N, K):
     scale = (torch.rand(K) * 1.5 + 0.25).cuda()
     shift = torch.randn(K).cuda() * 0.3
     C, ssum, ssq = _C.gemm_nt_fused(A, Bt, scale, shift, True, True)
+    ssum, ssq = ssum.sum(0), ssq.sum(0)  # slot-striped partials
     # oracle: transform A (with bf16 re-rounding, as the kernel does), GEMM
     Af = A.float()
     At = torch.relu(Af * scale[None, :] + shift[None, :]) \
@@ -641,6 +642,7 @@ def test_gemm_nt_fused_aout_materializes_transformed_a():
     shift = (torch.randn(K) * 0.3).cuda()
     C, ssum, ssq, aout = _C.gemm_nt_fused(A, Bt, scale, shift, True, True,
                                           True, None)
+    ssum, ssq = ssum.sum(0), ssq.sum(0)  # slot-striped partials
     ref_a = torch.relu(A.float() * scale[None, :] + shift[None, :]) \
         .to(torch.bfloat16)
     assert torch.equal(aout, ref_a)
