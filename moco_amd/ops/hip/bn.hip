@@ -128,34 +128,6 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
   }
 }
 
-// ---- finalize from slot-striped channel partials (conv-GEMM stats path) ---
-// The fused conv GEMM accumulates per-output-channel (sum, sumsq) of its
-// fp32 accumulators into a [P, C] slot-striped pair (P = 256 slots, see
-// conv1x1.hip STAT_SLOTS); this reduces the slots into mean/rstd and
-// updates the running stats — one tiny launch per conv.
-__global__ __launch_bounds__(256) void bn_finalize_kernel(
-    const float* __restrict__ ssum, const float* __restrict__ ssq,
-    float* __restrict__ mean, float* __restrict__ rstd,
-    float* __restrict__ running_mean, float* __restrict__ running_var,
-    int64_t R, int C, int P, float eps, float momentum) {
-  const int c = blockIdx.x * 256 + threadIdx.x;
-  if (c >= C) return;
-  float s = 0.f, q = 0.f;
-  for (int p = 0; p < P; ++p) {
-    s += ssum[(int64_t)p * C + c];
-    q += ssq[(int64_t)p * C + c];
-  }
-  const float m = s / (float)R;
-  const float var = fmaxf(q / (float)R - m * m, 0.f);
-  mean[c] = m;
-  rstd[c] = rsqrtf(var + eps);
-  if (running_mean != nullptr) {
-    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
-    const float unbiased = R > 1 ? var * (float)R / (float)(R - 1) : var;
-    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
-  }
-}
-
 // ---- forward apply --------------------------------------------------------
 template <bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_apply_kernel(
@@ -338,13 +310,17 @@ extern "C" int launch_bn_stats(void* stream, const void* x, float* part_s,
   return 0;
 }
 
+// Finalize mean/rstd (+ running stats) from the conv-GEMM epilogue's
+// slot-striped [P, C] partials.  Same math as the stats pass 2 above —
+// reuse its kernel (32 slot-groups x 8 channels per block through LDS;
+// a naive one-thread-per-channel slot loop measured 75 us per call).
 extern "C" int launch_bn_finalize(void* stream, const float* ssum,
                                   const float* ssq, float* mean, float* rstd,
                                   float* running_mean, float* running_var,
                                   int64_t R, int64_t C, int64_t P, float eps,
                                   float momentum) {
   hipStream_t s = (hipStream_t)stream;
-  bn_finalize_kernel<<<dim3((unsigned)((C + 255) / 256)), 256, 0, s>>>(
+  bn_stats_final_kernel<<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
       ssum, ssq, mean, rstd, running_mean, running_var, R, (int)C, (int)P,
       eps, momentum);
   HIP_CHECK_LAST();
