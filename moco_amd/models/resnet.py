@@ -116,6 +116,17 @@ class ResNet(nn.Module):
         self.avgpool = nn.AdaptiveAvgPool2d((1, 1))
         self.fc = nn.Linear(512 * block.expansion, num_classes)
 
+        # wire predecessor references across the trunk: a fused Bottleneck
+        # can take over its predecessor's bn3 backward-reduce (ops/block.py
+        # cross-block dgrad-epilogue fusion)
+        if block is Bottleneck:
+            trunk = [b for layer in (self.layer1, self.layer2, self.layer3,
+                                     self.layer4) for b in layer]
+            for prev_b, b in zip(trunk[:-1], trunk[1:]):
+                # bypass nn.Module.__setattr__: a plain reference, NOT a
+                # registered child module (state_dict must not see it)
+                object.__setattr__(b, "_prev_bottleneck", prev_b)
+
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
                 nn.init.kaiming_normal_(m.weight, mode="fan_out",

@@ -465,13 +465,18 @@ std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
                                              torch::Tensor Bt,
                                              torch::Tensor y, torch::Tensor z,
                                              torch::Tensor mean,
-                                             torch::Tensor rstd) {
+                                             torch::Tensor rstd,
+                                             c10::optional<torch::Tensor>
+                                                 c_add) {
   CHECK_CUDA_CONTIG(A);
   CHECK_CUDA_CONTIG(Bt);
   TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && is_bf16(y) && is_bf16(z));
   const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
   TORCH_CHECK(y.numel() == M * N && z.numel() == M * N);
   TORCH_CHECK(mean.numel() == N && rstd.numel() == N);
+  if (c_add.has_value())
+    TORCH_CHECK(is_bf16(*c_add) && c_add->is_contiguous() &&
+                c_add->numel() == M * N);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
   // slot-striped partials (conv1x1.hip STAT_SLOTS), reduced here
@@ -481,7 +486,8 @@ std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
       cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
       y.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), dbeta.data_ptr<float>(),
-      dgamma.data_ptr<float>()));
+      dgamma.data_ptr<float>(),
+      c_add.has_value() ? c_add->data_ptr() : nullptr));
   return {C, dbeta.sum(0), dgamma.sum(0)};
 }
 
@@ -572,6 +578,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("c_add") = c10::nullopt);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv1x1_strided", &conv1x1_strided);
-  mod.def("gemm_nt_dgrad_epi", &gemm_nt_dgrad_epi);
+  mod.def("gemm_nt_dgrad_epi", &gemm_nt_dgrad_epi, py::arg("A"),
+          py::arg("Bt"), py::arg("y"), py::arg("z"), py::arg("mean"),
+          py::arg("rstd"), py::arg("c_add") = c10::nullopt);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

@@ -91,9 +91,20 @@ def _conv1x1_wgrad(dz2d: torch.Tensor, x2d: torch.Tensor, nhw,
 class _FusedBottleneck(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, identity, w1, gamma1, beta1, w2, gamma2, beta2,
-                w3, gamma3, beta3, bn_state, conv2_meta, need_grad):
+                w3, gamma3, beta3, bn_state, conv2_meta, need_grad,
+                prev_bn3, prev_hand, state_sink, my_hand):
         # bn_state: (rm1, rv1, rm2, rv2, rm3, rv3, eps3tuple, mom)
         # conv2_meta: (stride, groups)
+        # prev_bn3: (z3_prev, mean3_prev, rstd3_prev) of the PRECEDING
+        #   bottleneck, when this block may fuse that block's bn3
+        #   backward-reduce + relu mask into its conv1 dgrad epilogue
+        #   (only identity-residual blocks qualify — see fused_bottleneck)
+        # prev_hand: the preceding block's handoff dict — this block's
+        #   backward deposits (dbeta3, dgamma3) there
+        # state_sink: dict this forward fills with (y, z3, mean3, rstd3)
+        #   for the NEXT block's dispatcher
+        # my_hand: this block's own handoff dict — its backward checks it
+        #   for a deposit from the next block
         rm1, rv1, rm2, rv2, rm3, rv3, eps, mom = bn_state
         stride, groups = conv2_meta
         N, Ci, H, W = x.shape
@@ -137,29 +148,54 @@ class _FusedBottleneck(torch.autograd.Function):
 
         ident2d = _as2d(identity if identity is not None else x)
         y = _C.bn_fwd_eval(z3, ident2d, gamma3, beta3, mean3, rstd3, True)
+        y4d = _as4d(y, N, Ho, Wo)
 
+        if prev_bn3 is not None:
+            z3p, mean3p, rstd3p = prev_bn3
+        else:
+            z3p = mean3p = rstd3p = x  # placeholders (unused)
         ctx.save_for_backward(x, z1, a1, z2, a2 if a2 is not None else z2,
                               z3, y, w1b, w2b, w3b, gamma1, gamma2, gamma3,
                               mean1, rstd1, mean2, rstd2, mean3, rstd3,
-                              scale2, shift2)
+                              scale2, shift2, z3p, mean3p, rstd3p)
         ctx.meta = (N, H, W, Ho, Wo, stride, groups,
-                    identity is not None, w1.stride(), w3.stride())
-        return _as4d(y, N, Ho, Wo)
+                    identity is not None, w1.stride(), w3.stride(),
+                    prev_bn3 is not None)
+        ctx.prev_hand = prev_hand
+        ctx.my_hand = my_hand
+        if state_sink is not None:
+            state_sink["y"] = y4d
+            state_sink["z3"] = z3
+            state_sink["mean3"] = mean3
+            state_sink["rstd3"] = rstd3
+        return y4d
 
     @staticmethod
     def backward(ctx, gy):
         (x, z1, a1, z2, a2, z3, y, w1b, w2b, w3b, gamma1, gamma2, gamma3,
-         mean1, rstd1, mean2, rstd2, mean3, rstd3, scale2,
-         shift2) = ctx.saved_tensors
+         mean1, rstd1, mean2, rstd2, mean3, rstd3, scale2, shift2,
+         z3p, mean3p, rstd3p) = ctx.saved_tensors
         (N, H, W, Ho, Wo, stride, groups, has_sep_identity, w1_strides,
-         w3_strides) = ctx.meta
+         w3_strides, has_prev) = ctx.meta
         _C = ext()
 
         gy2d = _as2d(gy.contiguous(memory_format=torch.channels_last))
 
-        # bn3 backward (full: relu mask, residual grad, stats terms)
-        dz3, dgamma3, dbeta3, dres = _C.bn_bwd(gy2d, y, z3, mean3, rstd3,
-                                               gamma3, True, True, True)
+        # bn3 backward.  When the NEXT block's conv1-dgrad epilogue already
+        # applied the relu mask and reduced dbeta3/dgamma3 (deposit in
+        # my_hand), only the dx pass remains and dres IS the incoming
+        # (pre-masked) grad; otherwise the full reduce + dx.
+        deposit = ctx.my_hand.pop("deposit", None) \
+            if ctx.my_hand is not None else None
+        if deposit is not None:
+            dbeta3, dgamma3 = deposit
+            dz3 = _C.bn_bwd_dx_only(gy2d, z3, mean3, rstd3, gamma3,
+                                    dbeta3, dgamma3)
+            dres = gy2d
+        else:
+            dz3, dgamma3, dbeta3, dres = _C.bn_bwd(gy2d, y, z3, mean3,
+                                                   rstd3, gamma3, True,
+                                                   True, True)
 
         # conv3 dgrad + fused bn2 backward-reduce (mask from a2, zhat
         # from z2) — gp2 is the relu-masked da2
@@ -185,19 +221,29 @@ class _FusedBottleneck(torch.autograd.Function):
         dz1, dgamma1, dbeta1 = _C.bn_bwd(da1, a1, z1, mean1, rstd1, gamma1,
                                          True, True, False)
 
-        # conv1 dgrad; the residual grad is folded into the store tail when
-        # the identity IS x (no downsample) — kills the separate add pass
+        # conv1 dgrad.  Identity-residual blocks fold the residual grad
+        # into the store tail (c_add); when the PRECEDING block is a fused
+        # bottleneck, the same epilogue also applies ITS bn3 relu mask
+        # (from x = that block's output) and reduces its dbeta3/dgamma3
+        # (deposited into prev_hand), so the grad handed upstream is
+        # already masked and that block skips its whole reduce pass.
         w1t = w1b.t().contiguous()
         x2d = _as2d(x)
-        dx2d = _C.gemm_nt_bf16(dz1, w1t,
-                               None if has_sep_identity else dres)
+        if has_prev and not has_sep_identity and ctx.prev_hand is not None:
+            dx2d, db3p, dg3p = _C.gemm_nt_dgrad_epi(
+                dz1, w1t, x2d, z3p, mean3p, rstd3p, dres)
+            ctx.prev_hand["deposit"] = (db3p, dg3p)
+        else:
+            dx2d = _C.gemm_nt_bf16(dz1, w1t,
+                                   None if has_sep_identity else dres)
         dw1 = _conv1x1_wgrad(dz1, x2d, (N, H, W), w1_strides)
 
         dx = _as4d(dx2d, N, H, W)
         dident = _as4d(dres, N, Ho, Wo) if has_sep_identity else None
 
         return (dx, dident, dw1, dgamma1, dbeta1, dw2.float(), dgamma2,
-                dbeta2, dw3, dgamma3, dbeta3, None, None, None)
+                dbeta2, dw3, dgamma3, dbeta3, None, None, None, None, None,
+                None, None)
 
 
 def _pow2_channels_ok(c: int) -> bool:
@@ -248,8 +294,28 @@ def fused_bottleneck(x: torch.Tensor, block) -> Optional[torch.Tensor]:
     conv2_meta = (block.conv2.stride[0], block.conv2.groups)
     need_grad = torch.is_grad_enabled() and (
         x.requires_grad or block.conv1.weight.requires_grad)
-    return _FusedBottleneck.apply(
+
+    # cross-block bn3-backward fusion: when the PRECEDING bottleneck ran
+    # fused this step (its state sink's output IS our input) and we have
+    # an identity residual, our conv1-dgrad epilogue can take over its
+    # bn3 reduce + relu mask.
+    prev_bn3 = prev_hand = None
+    prev = getattr(block, "_prev_bottleneck", None)
+    if (need_grad and identity is None and prev is not None):
+        st = getattr(prev, "_bn3_fwd_state", None)
+        if st is not None and st.get("y") is x:
+            prev_bn3 = (st["z3"], st["mean3"], st["rstd3"])
+            prev_hand = getattr(prev, "_bn3_hand", None)
+            prev._bn3_fwd_state = None  # consumed; don't retain activations
+    state_sink = {} if need_grad else None
+    my_hand = {} if need_grad else None
+    out = _FusedBottleneck.apply(
         x, identity, block.conv1.weight, block.bn1.weight, block.bn1.bias,
         block.conv2.weight, block.bn2.weight, block.bn2.bias,
         block.conv3.weight, block.bn3.weight, block.bn3.bias,
-        bn_state, conv2_meta, need_grad)
+        bn_state, conv2_meta, need_grad, prev_bn3, prev_hand, state_sink,
+        my_hand)
+    if need_grad:
+        block._bn3_fwd_state = state_sink
+        block._bn3_hand = my_hand
+    return out

@@ -381,13 +381,17 @@ extern "C" int launch_gemm_nt_fused_bf16(void* stream, const void* A,
 // ---------------------------------------------------------------------------
 namespace {
 
+// c_add (nullable): epilogue C += c_add BEFORE the relu mask — the
+// residual-path gradient joins the dgrad sum, then the whole sum is masked
+// (y = relu(bn(z) + res): upstream grads add first, then the mask).
 template <int BNT>
 __global__ __launch_bounds__(256, 2) void gemm_nt_dgrad_epi_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ Bt,
     bf16* __restrict__ C, int64_t M, int N, int K,
     const bf16* __restrict__ y, const bf16* __restrict__ z,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ dbeta, float* __restrict__ dgamma) {
+    float* __restrict__ dbeta, float* __restrict__ dgamma,
+    const bf16* __restrict__ c_add) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   constexpr int MR = BNT == 128 ? 4 : 2;
   constexpr int ABYTES = BM * BK * 2;
@@ -477,10 +481,14 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_dgrad_epi_kernel(
     const bf16x8 gv = *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
     const bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + off);
     const bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + off);
+    bf16x8 av = {};
+    if (c_add != nullptr)
+      av = *reinterpret_cast<const bf16x8*>(c_add + off);
     bf16x8 gp;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float g = (float)gv[j];
+      if (c_add != nullptr) g += (float)av[j];
       if ((float)yv[j] <= 0.f) g = 0.f;
       gp[j] = (__bf16)g;
       sdb[j] += g;
@@ -515,7 +523,7 @@ extern "C" int launch_gemm_nt_dgrad_epi(void* stream, const void* A,
                                         int64_t N, int64_t K, const void* y,
                                         const void* z, const float* mean,
                                         const float* rstd, float* dbeta,
-                                        float* dgamma) {
+                                        float* dgamma, const void* c_add) {
   if (M % 128 != 0 || K % 64 != 0 || N % 64 != 0)
     return (int)hipErrorInvalidValue;
   hipStream_t s = (hipStream_t)stream;
@@ -525,11 +533,13 @@ extern "C" int launch_gemm_nt_dgrad_epi(void* stream, const void* A,
   if (BNT == 128)
     gemm_nt_dgrad_epi_kernel<128><<<grid, 256, smem, s>>>(
         (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,
-        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma);
+        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma,
+        (const bf16*)c_add);
   else
     gemm_nt_dgrad_epi_kernel<64><<<grid, 256, smem, s>>>(
         (const bf16*)A, (const bf16*)Bt, (bf16*)C, M, (int)N, (int)K,
-        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma);
+        (const bf16*)y, (const bf16*)z, mean, rstd, dbeta, dgamma,
+        (const bf16*)c_add);
   HIP_CHECK_LAST();
   return 0;
 }

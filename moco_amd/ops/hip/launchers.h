@@ -153,4 +153,5 @@ extern "C"
 int launch_gemm_nt_dgrad_epi(void* stream, const void* A, const void* Bt,
                              void* C, int64_t M, int64_t N, int64_t K,
                              const void* y, const void* z, const float* mean,
-                             const float* rstd, float* dbeta, float* dgamma);
+                             const float* rstd, float* dbeta, float* dgamma,
+                             const void* c_add);

@@ -174,3 +174,38 @@ def test_fused_bottleneck_fallback_on_odd_shapes():
     with torch.autocast("cuda", dtype=torch.bfloat16):
         y = blk(x)  # generic path
     assert y.shape == (3, 256, 7, 7)
+
+
+def test_fused_bottleneck_chain_handoff():
+    """Two stacked identity-residual blocks: block 2's conv1-dgrad epilogue
+    takes over block 1's bn3 backward-reduce (cross-block fusion).  Checked
+    against the fp32 oracle like the single-block test."""
+    inplanes, planes, N, H = 256, 64, 2, 8
+    torch.manual_seed(41)
+    b1 = _make_block(inplanes, planes, seed=41)
+    b2 = _make_block(inplanes, planes, seed=42)
+    assert b1.downsample is None and b2.downsample is None
+    chain = torch.nn.Sequential(b1, b2).cuda().train() \
+        .to(memory_format=torch.channels_last)
+    object.__setattr__(chain[1], "_prev_bottleneck", chain[0])
+    chain_gen = copy.deepcopy(chain)
+    chain_f32 = copy.deepcopy(chain)
+
+    torch.manual_seed(8)
+    x32 = (torch.randn(N, inplanes, H, H) * 0.5).cuda() \
+        .contiguous(memory_format=torch.channels_last)
+    xb = x32.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+
+    y_f, g_f = _run(chain, xb, "fused")
+    # handoff engaged: block 2 consumed block 1's forward state
+    assert chain[0]._bn3_fwd_state is None
+    y_g, g_g = _run(chain_gen, xb, "generic")
+    y_o, g_o = _run(chain_f32, x32, "fp32")
+
+    ef, eg = float((y_f - y_o).abs().max()), float((y_g - y_o).abs().max())
+    assert ef <= 2.0 * eg + 1e-2, (ef, eg)
+    for name in g_o:
+        e1 = float((g_f[name] - g_o[name]).abs().max())
+        e2 = float((g_g[name] - g_o[name]).abs().max())
+        floor = 1e-2 * max(1.0, float(g_o[name].abs().max()))
+        assert e1 <= 2.0 * e2 + floor, (name, e1, e2)
