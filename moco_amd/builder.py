@@ -70,6 +70,14 @@ class MoCo(nn.Module):
         self._queue_kc: Optional[torch.Tensor] = None  # [K, dim] bf16 (GPU)
         self._ptr_host: Optional[int] = None  # avoids a per-step device sync
         self._step = 0
+        # hipGraph-replay-safe stepping (set by the engine): enqueue is
+        # deferred to flush_queue() after backward (no queue snapshot
+        # needed; ring ptr advanced on device), and at world size 1 the
+        # shuffling-BN exchange is skipped — a permutation is a no-op for
+        # batch statistics and is exactly inverted by the unshuffle, so
+        # single-rank semantics are unchanged (FP reduction order aside).
+        self.graph_safe = False
+        self._pending_keys: Optional[torch.Tensor] = None
 
     # -- key-encoder EMA ----------------------------------------------------
     @torch.no_grad()
@@ -103,11 +111,27 @@ class MoCo(nn.Module):
     def _dequeue_and_enqueue(self, keys: torch.Tensor):
         keys = parallel.concat_all_gather(keys)  # C6: all ranks enqueue all
         qkc = self._queue_kc if self.queue.is_cuda else None
+        if self.graph_safe and self.queue.is_cuda:
+            # device-side ring position: correct under hipGraph replay
+            from .ops import ext
+            ext().queue_enqueue_dev(self.queue, qkc, keys.contiguous(),
+                                    self.queue_ptr)
+            self._ptr_host = None  # host mirror invalid in graph mode
+            return
         if self._ptr_host is None:
             self._ptr_host = int(self.queue_ptr.item())
         queue_enqueue_(self.queue, qkc, self.queue_ptr, keys,
                        ptr=self._ptr_host)
         self._ptr_host = (self._ptr_host + keys.shape[0]) % self.K
+
+    @torch.no_grad()
+    def flush_queue(self):
+        """Perform the enqueue deferred by a graph-safe forward (call after
+        backward; the queue must not change before the contrastive
+        backward reads it)."""
+        if self._pending_keys is not None:
+            self._dequeue_and_enqueue(self._pending_keys)
+            self._pending_keys = None
 
     # -- shuffling-BN (world-aware; see parallel/comm.py) -------------------
     @torch.no_grad()
@@ -129,17 +153,19 @@ class MoCo(nn.Module):
         if step is None:
             step = self._step
             self._step += 1
+        graph_safe = self.graph_safe and parallel.get_world_size() == 1
         # the shuffle permutation is derived from (seed, step) on every rank
         # independently — verify the counters agree periodically (a desynced
         # rank would otherwise corrupt the key pairing SILENTLY)
-        if step % 64 == 0:
+        if step % 64 == 0 and not graph_safe:
             parallel.assert_step_sync(step, im_q.device)
 
         with torch.no_grad():
             # start the key-batch exchange on the comm stream FIRST so the
             # xGMI transfer overlaps the query forward (join at key forward)
-            shuffle_job = parallel.begin_shuffle(im_k, step,
-                                                 self.shuffle_seed)
+            if not graph_safe:
+                shuffle_job = parallel.begin_shuffle(im_k, step,
+                                                     self.shuffle_seed)
             # EMA before the key forward, once per step (reference invariant)
             self._momentum_update_key_encoder()
 
@@ -147,10 +173,17 @@ class MoCo(nn.Module):
         q = l2_normalize(q, dim=1)
 
         with torch.no_grad():
-            im_k_sh, idx_unshuffle = shuffle_job.wait()
-            k = self.encoder_k(im_k_sh)
-            k = l2_normalize(k, dim=1)
-            k = self._batch_unshuffle(k, idx_unshuffle)
+            if graph_safe:
+                # world-1 shuffle is exactly inverted by the unshuffle and
+                # batch statistics are permutation-invariant: skip it (the
+                # host-generated per-step permutation is not replay-safe)
+                k = self.encoder_k(im_k)
+                k = l2_normalize(k, dim=1)
+            else:
+                im_k_sh, idx_unshuffle = shuffle_job.wait()
+                k = self.encoder_k(im_k_sh)
+                k = l2_normalize(k, dim=1)
+                k = self._batch_unshuffle(k, idx_unshuffle)
 
         if q.is_cuda and (q.dtype == torch.bfloat16
                           or torch.is_autocast_enabled()):
@@ -158,6 +191,15 @@ class MoCo(nn.Module):
             # takes the hipBLASLt fallback inside contrastive_logits
             q = q.to(torch.bfloat16)
             k = k.to(torch.bfloat16)
+        if graph_safe:
+            # enqueue deferred past backward: the queue is stable for the
+            # whole fwd+bwd — no snapshot at all
+            logits = contrastive_logits(q, k, self._working_queue(),
+                                        self.queue, self.T, stable=True)
+            labels = torch.zeros(logits.shape[0], dtype=torch.long,
+                                 device=logits.device)
+            self._pending_keys = k.detach()
+            return logits, labels
         # tell the op which queue block the enqueue below will overwrite so
         # its backward snapshots only those columns (not the whole ring)
         if self._ptr_host is None:

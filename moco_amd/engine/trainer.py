@@ -30,7 +30,8 @@ class MoCoStepper:
                  aug_plus: bool = False, out_size: int = 224,
                  dtype: torch.dtype = torch.bfloat16,
                  channels_last: bool = True,
-                 aug_generator: Optional[torch.Generator] = None):
+                 aug_generator: Optional[torch.Generator] = None,
+                 use_graph: Optional[bool] = None):
         self.model = model
         self.optimizer = optimizer
         self.device = device
@@ -48,6 +49,19 @@ class MoCoStepper:
         self.aug_generator = aug_generator
         self._timing = False
         self._phase_events = None
+        # hipGraph capture of the whole step (fwd+bwd+opt+enqueue): kills
+        # the per-step launch/dispatch host cost.  World-1 only (RCCL
+        # collectives stay eager) and off while phase-timing.
+        if use_graph is None:
+            import os
+            use_graph = (self.is_cuda
+                         and os.environ.get("MOCO_AMD_NO_GRAPH", "0") != "1"
+                         and not (torch.distributed.is_available()
+                                  and torch.distributed.is_initialized()))
+        self.use_graph = bool(use_graph) and self.is_cuda
+        self._graph = None
+        self._graph_key = None
+        self._gwarm = 0
 
     def augment(self, raw: torch.Tensor):
         raw = raw.to(self.device, non_blocking=True)
@@ -60,6 +74,12 @@ class MoCoStepper:
     def step(self, im_q: torch.Tensor, im_k: torch.Tensor):
         """One optimization step on pre-augmented crops.
         Returns (loss, logits, labels)."""
+        raw_model = getattr(self.model, "module", self.model)
+        if getattr(raw_model, "graph_safe", False):
+            # leaving graph mode (e.g. a phase-timing step): back to the
+            # eager enqueue-in-forward path
+            raw_model.flush_queue()
+            raw_model.graph_safe = False
         autocast_on = self.is_cuda and self.dtype != torch.float32
         self._mark("fwd")
         with torch.autocast("cuda", dtype=self.dtype, enabled=autocast_on):
@@ -78,7 +98,63 @@ class MoCoStepper:
         self._phase_events = [] if self._timing else None
         self._mark("aug")
         im_q, im_k = self.augment(raw)
+        if self.use_graph and not self._timing:
+            return self.step_graphed(im_q, im_k)
         return self.step(im_q, im_k)
+
+    # -- hipGraph whole-step capture ----------------------------------------
+    def _inner_step(self, im_q, im_k):
+        """The capturable step body: zero -> fwd -> loss -> bwd -> opt ->
+        deferred enqueue.  Grads persist (set_to_none=False) so their
+        addresses are stable across capture/replay."""
+        self.optimizer.zero_grad(set_to_none=False)
+        with torch.autocast("cuda", dtype=self.dtype,
+                            enabled=self.dtype != torch.float32):
+            logits, labels = self.model(im_q, im_k)
+        loss = moco_loss(logits)
+        loss.backward()
+        self.optimizer.step()
+        raw_model = getattr(self.model, "module", self.model)
+        raw_model.flush_queue()
+        return loss, logits, labels
+
+    def step_graphed(self, im_q: torch.Tensor, im_k: torch.Tensor):
+        key = (im_q.shape, self.optimizer.param_groups[0]["lr"])
+        if self._graph_key != key:
+            self._graph = None
+            self._gwarm = 0
+            self._graph_key = key
+        raw_model = getattr(self.model, "module", self.model)
+        raw_model.graph_safe = True
+        if self._graph is None:
+            if self._gwarm == 0:
+                self._gq = im_q.clone()
+                self._gk = im_k.clone()
+            else:
+                self._gq.copy_(im_q)
+                self._gk.copy_(im_k)
+            if self._gwarm < 3:
+                # warmup on a side stream (graph-capture prerequisite);
+                # also builds the SGD/EMA chunk tables and MIOpen finds at
+                # stable addresses
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    out = self._inner_step(self._gq, self._gk)
+                torch.cuda.current_stream().wait_stream(s)
+                self._gwarm += 1
+                return out
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._gout = self._inner_step(self._gq, self._gk)
+            self._graph = g
+            # capture itself executes nothing: replay once to take the step
+            g.replay()
+            return self._gout
+        self._gq.copy_(im_q)
+        self._gk.copy_(im_k)
+        self._graph.replay()
+        return self._gout
 
     # -- optional per-phase GPU timing (HIP events; SURVEY §5.1) ------------
     def enable_timing(self, on: bool = True):

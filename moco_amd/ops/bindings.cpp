@@ -123,6 +123,31 @@ torch::Tensor softmax_ce0_bwd(torch::Tensor logits, torch::Tensor row_max,
 }
 
 // ---------------------------------------------------------------------------
+// graph-replay-safe enqueue: ring position read/advanced from the queue_ptr
+// buffer on device (a host ptr would be frozen into a captured graph)
+void queue_enqueue_dev(torch::Tensor queue,
+                       c10::optional<torch::Tensor> queue_kc,
+                       torch::Tensor keys, torch::Tensor queue_ptr) {
+  CHECK_CUDA_CONTIG(queue);
+  CHECK_CUDA_CONTIG(keys);
+  TORCH_CHECK(queue.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(queue_ptr.is_cuda() &&
+              queue_ptr.scalar_type() == torch::kInt64);
+  const int64_t C = queue.size(0), K = queue.size(1), B = keys.size(0);
+  TORCH_CHECK(keys.size(1) == C && K % B == 0);
+  void* kc = nullptr;
+  if (queue_kc.has_value()) {
+    CHECK_CUDA_CONTIG(*queue_kc);
+    TORCH_CHECK(is_bf16(*queue_kc) && queue_kc->size(0) == K &&
+                queue_kc->size(1) == C);
+    kc = queue_kc->data_ptr();
+  }
+  at::cuda::CUDAGuard guard(queue.device());
+  CHECK_LAUNCH(launch_queue_enqueue_dev(
+      cur_stream(), queue.data_ptr<float>(), kc, keys.data_ptr(),
+      is_bf16(keys), C, K, B, queue_ptr.data_ptr<int64_t>()));
+}
+
 void queue_enqueue(torch::Tensor queue, c10::optional<torch::Tensor> queue_kc,
                    torch::Tensor keys, int64_t ptr) {
   CHECK_CUDA_CONTIG(queue);
@@ -559,6 +584,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("softmax_ce0_fwd", &softmax_ce0_fwd);
   mod.def("softmax_ce0_bwd", &softmax_ce0_bwd);
   mod.def("queue_enqueue", &queue_enqueue);
+  mod.def("queue_enqueue_dev", &queue_enqueue_dev);
   mod.def("multi_tensor_ema", &multi_tensor_ema);
   mod.def("multi_tensor_sgd", &multi_tensor_sgd);
   mod.def("ema_flat", &ema_flat);

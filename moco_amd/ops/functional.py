@@ -85,7 +85,8 @@ class _ContrastiveLogits(torch.autograd.Function):
     def forward(ctx, q: torch.Tensor, k: torch.Tensor,
                 queue_kc: torch.Tensor, queue_ck: torch.Tensor,
                 T: float, snap_ptr: Optional[int] = None,
-                snap_span: Optional[int] = None) -> torch.Tensor:
+                snap_span: Optional[int] = None,
+                stable: bool = False) -> torch.Tensor:
         q = q.contiguous()
         k = k.contiguous()
         # MFMA kernel constraints (non-standard dims run the hipBLASLt
@@ -115,7 +116,13 @@ class _ContrastiveLogits(torch.autograd.Function):
         # (C x global-batch fp32, ~131 KB at the reference config, vs 33 MB
         # for the clone at K=65536) and splice them back around the backward
         # GEMM.  Without that info (direct op-level use), full clone.
-        if snap_ptr is not None and snap_span is not None:
+        if stable:
+            # caller guarantees the queue is NOT modified before backward
+            # (deferred enqueue / hipGraph mode): no snapshot at all
+            ctx.save_for_backward(k)
+            ctx.queue_ck_live = queue_ck
+            ctx.snap = "stable"
+        elif snap_ptr is not None and snap_span is not None:
             cols = queue_ck[:, snap_ptr:snap_ptr + snap_span].clone()
             ctx.save_for_backward(k, cols)
             ctx.queue_ck_live = queue_ck  # plain attr: bypasses the version
@@ -130,8 +137,13 @@ class _ContrastiveLogits(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_logits: torch.Tensor):
-        k, saved = ctx.saved_tensors
-        if ctx.snap is not None:
+        if ctx.snap == "stable":
+            (k,) = ctx.saved_tensors
+            saved = None
+            queue_ck = ctx.queue_ck_live
+        else:
+            k, saved = ctx.saved_tensors
+        if ctx.snap is not None and ctx.snap != "stable":
             ptr, span = ctx.snap
             queue_ck = ctx.queue_ck_live
             # splice the forward-time columns back in for the GEMM, then
@@ -139,7 +151,7 @@ class _ContrastiveLogits(torch.autograd.Function):
             # on the forward's stream before the next step's enqueue.
             cur = queue_ck[:, ptr:ptr + span].clone()
             queue_ck[:, ptr:ptr + span] = saved
-        else:
+        elif ctx.snap is None:
             queue_ck = saved
         if use_hip(grad_logits) and ctx.shapes_ok:
             gq = ext().contrastive_bwd(grad_logits.contiguous(), k,
@@ -148,17 +160,19 @@ class _ContrastiveLogits(torch.autograd.Function):
             g = grad_logits.float()
             gq = (g[:, :1] * k.float() +
                   g[:, 1:] @ queue_ck.float().t()) * ctx.inv_T
-        if ctx.snap is not None:
+        if ctx.snap is not None and ctx.snap != "stable":
             queue_ck[:, ptr:ptr + span] = cur
-        return gq.to(ctx.q_dtype), None, None, None, None, None, None
+        return (gq.to(ctx.q_dtype), None, None, None, None, None, None,
+                None)
 
 
 def contrastive_logits(q: torch.Tensor, k: torch.Tensor,
                        queue_kc: torch.Tensor, queue_ck: torch.Tensor,
                        T: float, snap_ptr: Optional[int] = None,
-                       snap_span: Optional[int] = None) -> torch.Tensor:
+                       snap_span: Optional[int] = None,
+                       stable: bool = False) -> torch.Tensor:
     return _ContrastiveLogits.apply(q, k, queue_kc, queue_ck, T, snap_ptr,
-                                    snap_span)
+                                    snap_span, stable)
 
 
 # ---------------------------------------------------------------------------

@@ -39,6 +39,10 @@ int launch_softmax_ce0_bwd(void* stream, const float* logits,
 int launch_queue_enqueue(void* stream, float* queue, void* queue_kc,
                          const void* keys, bool keys_bf16, int64_t C,
                          int64_t K, int64_t B, int64_t ptr);
+// graph-replay-safe variant: ring position read/advanced from ptr_dev
+int launch_queue_enqueue_dev(void* stream, float* queue, void* queue_kc,
+                             const void* keys, bool keys_bf16, int64_t C,
+                             int64_t K, int64_t B, int64_t* ptr_dev);
 
 // ---- multi_tensor.hip -----------------------------------------------------
 int launch_multi_tensor_ema(void* stream, const int64_t* ptrs_dst,
