@@ -133,11 +133,15 @@ class MoCoStepper:
             else:
                 self._gq.copy_(im_q)
                 self._gk.copy_(im_k)
+            if self._gwarm == 0:
+                self._gstream = torch.cuda.Stream()
             if self._gwarm < 3:
-                # warmup on a side stream (graph-capture prerequisite);
-                # also builds the SGD/EMA chunk tables and MIOpen finds at
-                # stable addresses
-                s = torch.cuda.Stream()
+                # warmup on the side stream that will also host the capture
+                # (AccumulateGrad nodes pin the stream they were created
+                # on; capturing on a different stream trips cross-stream
+                # event syncs) — also builds the SGD/EMA chunk tables and
+                # MIOpen finds at stable addresses
+                s = self._gstream
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
                     out = self._inner_step(self._gq, self._gk)
@@ -145,7 +149,7 @@ class MoCoStepper:
                 self._gwarm += 1
                 return out
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with torch.cuda.graph(g, stream=self._gstream):
                 self._gout = self._inner_step(self._gq, self._gk)
             self._graph = g
             # capture itself executes nothing: replay once to take the step

@@ -110,15 +110,25 @@ std::vector<torch::Tensor> softmax_ce0_fwd(torch::Tensor logits) {
   return {loss_rows, row_max, row_lse};
 }
 
+// gscale (optional fp32 scalar tensor): extra scale read on device — the
+// upstream grad under hipGraph capture, where a host .item() would sync
 torch::Tensor softmax_ce0_bwd(torch::Tensor logits, torch::Tensor row_max,
-                              torch::Tensor row_lse, double scale) {
+                              torch::Tensor row_lse, double scale,
+                              c10::optional<torch::Tensor> gscale) {
   CHECK_CUDA_CONTIG(logits);
   at::cuda::CUDAGuard guard(logits.device());
   auto dlogits = torch::empty_like(logits);
+  const float* gs = nullptr;
+  if (gscale.has_value()) {
+    TORCH_CHECK(gscale->is_cuda() &&
+                gscale->scalar_type() == torch::kFloat32 &&
+                gscale->numel() == 1);
+    gs = gscale->data_ptr<float>();
+  }
   CHECK_LAUNCH(launch_softmax_ce0_bwd(
       cur_stream(), logits.data_ptr<float>(), row_max.data_ptr<float>(),
       row_lse.data_ptr<float>(), dlogits.data_ptr<float>(), (float)scale,
-      logits.size(0), logits.size(1)));
+      gs, logits.size(0), logits.size(1)));
   return dlogits;
 }
 
@@ -582,7 +592,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("contrastive_fwd", &contrastive_fwd);
   mod.def("contrastive_bwd", &contrastive_bwd);
   mod.def("softmax_ce0_fwd", &softmax_ce0_fwd);
-  mod.def("softmax_ce0_bwd", &softmax_ce0_bwd);
+  mod.def("softmax_ce0_bwd", &softmax_ce0_bwd, py::arg("logits"),
+          py::arg("row_max"), py::arg("row_lse"), py::arg("scale"),
+          py::arg("gscale") = c10::nullopt);
   mod.def("queue_enqueue", &queue_enqueue);
   mod.def("queue_enqueue_dev", &queue_enqueue_dev);
   mod.def("multi_tensor_ema", &multi_tensor_ema);

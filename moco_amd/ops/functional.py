@@ -198,8 +198,10 @@ class _SoftmaxCELabel0(torch.autograd.Function):
         logits, row_max, row_lse = ctx.saved_tensors
         n = logits.shape[0]
         if use_hip(logits):
+            # grad_loss stays on device (a host read would sync — and is
+            # illegal under hipGraph capture); the kernel multiplies it in
             dlogits = ext().softmax_ce0_bwd(logits, row_max, row_lse,
-                                            float(grad_loss) / n)
+                                            1.0 / n, grad_loss.float())
         else:
             lf = logits.float()
             p = torch.exp(lf - (row_max + row_lse)[:, None])

@@ -32,7 +32,8 @@ int launch_softmax_ce0_fwd(void* stream, const float* logits, float* loss_rows,
                            int64_t M);
 int launch_softmax_ce0_bwd(void* stream, const float* logits,
                            const float* row_max, const float* row_lse,
-                           float* dlogits, float scale, int64_t N, int64_t M);
+                           float* dlogits, float scale, const float* gscale,
+                           int64_t N, int64_t M);
 
 // ---- queue.hip ------------------------------------------------------------
 // queue: [C,K] fp32; queue_kc: [K,C] bf16 (nullable); keys: [B,C]

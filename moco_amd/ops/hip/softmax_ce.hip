@@ -71,13 +71,16 @@ __global__ __launch_bounds__(256) void ce0_fwd_kernel(
   }
 }
 
+// gscale (nullable): extra scale read FROM DEVICE (the upstream grad of
+// the mean loss under hipGraph capture, where a host read would sync)
 __global__ __launch_bounds__(256) void ce0_bwd_kernel(
     const float* __restrict__ logits, const float* __restrict__ row_max,
     const float* __restrict__ row_lse, float* __restrict__ dlogits,
-    float scale, int64_t N, int64_t M) {
+    float scale, const float* __restrict__ gscale, int64_t N, int64_t M) {
   const int64_t idx = ((int64_t)blockIdx.x * 256 + threadIdx.x) * 4;
   const int64_t total = N * M;
   if (idx >= total) return;
+  if (gscale != nullptr) scale *= *gscale;
   const int64_t row = idx / M;
   const int64_t col = idx - row * M;
   const float z = row_max[row] + row_lse[row];
@@ -114,12 +117,13 @@ extern "C" int launch_softmax_ce0_fwd(void* stream, const float* logits,
 extern "C" int launch_softmax_ce0_bwd(void* stream, const float* logits,
                                       const float* row_max,
                                       const float* row_lse, float* dlogits,
-                                      float scale, int64_t N, int64_t M) {
+                                      float scale, const float* gscale,
+                                      int64_t N, int64_t M) {
   hipStream_t s = (hipStream_t)stream;
   const int64_t total = (N * M + 3) / 4;
   const int64_t blocks = (total + 255) / 256;
   ce0_bwd_kernel<<<dim3((unsigned)blocks), 256, 0, s>>>(
-      logits, row_max, row_lse, dlogits, scale, N, M);
+      logits, row_max, row_lse, dlogits, scale, gscale, N, M);
   HIP_CHECK_LAST();
   return 0;
 }
