@@ -104,22 +104,23 @@ class MoCoStepper:
 
     # -- hipGraph whole-step capture ----------------------------------------
     def _inner_step(self, im_q, im_k):
-        """The capturable step body: zero -> fwd -> loss -> bwd -> opt ->
-        deferred enqueue.  Grads persist (set_to_none=False) so their
-        addresses are stable across capture/replay."""
-        self.optimizer.zero_grad(set_to_none=False)
+        """The capturable step body: fwd -> loss -> bwd -> deferred
+        enqueue.  Grads are None on entry (set_to_none=True), so the
+        captured backward ASSIGNS each grad buffer — replays overwrite
+        them in place with no per-step zero or accumulate kernels.  The
+        fused SGD runs eagerly after each replay (one launch; grad
+        addresses are pool-stable, and lr changes need no re-capture)."""
         with torch.autocast("cuda", dtype=self.dtype,
                             enabled=self.dtype != torch.float32):
             logits, labels = self.model(im_q, im_k)
         loss = moco_loss(logits)
         loss.backward()
-        self.optimizer.step()
         raw_model = getattr(self.model, "module", self.model)
         raw_model.flush_queue()
         return loss, logits, labels
 
     def step_graphed(self, im_q: torch.Tensor, im_k: torch.Tensor):
-        key = (im_q.shape, self.optimizer.param_groups[0]["lr"])
+        key = im_q.shape
         if self._graph_key != key:
             self._graph = None
             self._gwarm = 0
@@ -130,34 +131,38 @@ class MoCoStepper:
             if self._gwarm == 0:
                 self._gq = im_q.clone()
                 self._gk = im_k.clone()
+                self._gstream = torch.cuda.Stream()
             else:
                 self._gq.copy_(im_q)
                 self._gk.copy_(im_k)
-            if self._gwarm == 0:
-                self._gstream = torch.cuda.Stream()
             if self._gwarm < 3:
                 # warmup on the side stream that will also host the capture
                 # (AccumulateGrad nodes pin the stream they were created
                 # on; capturing on a different stream trips cross-stream
-                # event syncs) — also builds the SGD/EMA chunk tables and
-                # MIOpen finds at stable addresses
+                # event syncs) — also runs the MIOpen finds and builds the
+                # EMA chunk tables at stable addresses
+                self.optimizer.zero_grad(set_to_none=True)
                 s = self._gstream
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
                     out = self._inner_step(self._gq, self._gk)
                 torch.cuda.current_stream().wait_stream(s)
+                self.optimizer.step()
                 self._gwarm += 1
                 return out
+            self.optimizer.zero_grad(set_to_none=True)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g, stream=self._gstream):
                 self._gout = self._inner_step(self._gq, self._gk)
             self._graph = g
             # capture itself executes nothing: replay once to take the step
             g.replay()
+            self.optimizer.step()
             return self._gout
         self._gq.copy_(im_q)
         self._gk.copy_(im_k)
         self._graph.replay()
+        self.optimizer.step()
         return self._gout
 
     # -- optional per-phase GPU timing (HIP events) ------------

@@ -325,7 +325,9 @@ torch::Tensor gemm_nt_bf16(torch::Tensor A, torch::Tensor Bt,
 std::vector<torch::Tensor> gemm_nt_fused(
     torch::Tensor A, torch::Tensor Bt, c10::optional<torch::Tensor> a_scale,
     c10::optional<torch::Tensor> a_shift, bool a_relu, bool want_stats,
-    bool want_aout, c10::optional<torch::Tensor> c_add) {
+    bool want_aout, c10::optional<torch::Tensor> c_add,
+    c10::optional<torch::Tensor> ws_sum,
+    c10::optional<torch::Tensor> ws_ssq) {
   CHECK_CUDA_CONTIG(A);
   CHECK_CUDA_CONTIG(Bt);
   TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && A.size(1) == Bt.size(1));
@@ -343,9 +345,18 @@ std::vector<torch::Tensor> gemm_nt_fused(
   torch::Tensor ssum, ssq, aout;
   if (want_stats) {
     // slot-striped [256, N] partials (see conv1x1.hip STAT_SLOTS);
-    // bn_finalize reduces the slots
-    ssum = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
-    ssq = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+    // bn_finalize reduces the slots.  When the caller provides a
+    // workspace pair it must be pre-zeroed — the finalize(zero_src=True)
+    // restores that invariant, so no fill kernels run per step.
+    if (ws_sum.has_value()) {
+      TORCH_CHECK(ws_sum->numel() == 256 * N && ws_sum->is_contiguous() &&
+                  ws_ssq->numel() == 256 * N && ws_ssq->is_contiguous());
+      ssum = *ws_sum;
+      ssq = *ws_ssq;
+    } else {
+      ssum = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+      ssq = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+    }
   }
   if (want_aout) aout = torch::empty({M, K}, A.options());
   CHECK_LAUNCH(launch_gemm_nt_fused_bf16(
@@ -398,7 +409,8 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
                                        int64_t R,
                                        c10::optional<torch::Tensor> rmean,
                                        c10::optional<torch::Tensor> rvar,
-                                       double eps, double momentum) {
+                                       double eps, double momentum,
+                                       bool zero_src) {
   CHECK_CUDA_CONTIG(ssum);
   CHECK_CUDA_CONTIG(ssq);
   const int64_t C = ssum.dim() == 2 ? ssum.size(1) : ssum.numel();
@@ -411,7 +423,7 @@ std::vector<torch::Tensor> bn_finalize(torch::Tensor ssum, torch::Tensor ssq,
       mean.data_ptr<float>(), rstd.data_ptr<float>(),
       rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
       rvar.has_value() ? rvar->data_ptr<float>() : nullptr, R, C, P,
-      (float)eps, (float)momentum));
+      (float)eps, (float)momentum, zero_src));
   return {mean, rstd};
 }
 
@@ -502,7 +514,11 @@ std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
                                              torch::Tensor mean,
                                              torch::Tensor rstd,
                                              c10::optional<torch::Tensor>
-                                                 c_add) {
+                                                 c_add,
+                                             c10::optional<torch::Tensor>
+                                                 ws_db,
+                                             c10::optional<torch::Tensor>
+                                                 ws_dg) {
   CHECK_CUDA_CONTIG(A);
   CHECK_CUDA_CONTIG(Bt);
   TORCH_CHECK(is_bf16(A) && is_bf16(Bt) && is_bf16(y) && is_bf16(z));
@@ -514,16 +530,30 @@ std::vector<torch::Tensor> gemm_nt_dgrad_epi(torch::Tensor A,
                 c_add->numel() == M * N);
   at::cuda::CUDAGuard guard(A.device());
   auto C = torch::empty({M, N}, A.options());
-  // slot-striped partials (conv1x1.hip STAT_SLOTS), reduced here
-  auto dbeta = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
-  auto dgamma = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+  // slot-striped partials (conv1x1.hip STAT_SLOTS), reduced by the
+  // self-zeroing slot_reduce2 (caller workspace must be pre-zeroed)
+  torch::Tensor dbeta, dgamma;
+  if (ws_db.has_value()) {
+    TORCH_CHECK(ws_db->numel() == 256 * N && ws_db->is_contiguous() &&
+                ws_dg->numel() == 256 * N && ws_dg->is_contiguous());
+    dbeta = *ws_db;
+    dgamma = *ws_dg;
+  } else {
+    dbeta = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+    dgamma = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+  }
   CHECK_LAUNCH(launch_gemm_nt_dgrad_epi(
       cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
       y.data_ptr(), z.data_ptr(), mean.data_ptr<float>(),
       rstd.data_ptr<float>(), dbeta.data_ptr<float>(),
       dgamma.data_ptr<float>(),
       c_add.has_value() ? c_add->data_ptr() : nullptr));
-  return {C, dbeta.sum(0), dgamma.sum(0)};
+  auto db = torch::empty({N}, A.options().dtype(torch::kFloat32));
+  auto dg = torch::empty({N}, A.options().dtype(torch::kFloat32));
+  CHECK_LAUNCH(launch_slot_reduce2(
+      cur_stream(), dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+      db.data_ptr<float>(), dg.data_ptr<float>(), N, 256));
+  return {C, db, dg};
 }
 
 // ---------------------------------------------------------------------------
@@ -605,7 +635,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bn_fwd_eval", &bn_fwd_eval);
   mod.def("bn_bwd", &bn_bwd);
   mod.def("bn_stats", &bn_stats);
-  mod.def("bn_finalize", &bn_finalize);
+  mod.def("bn_finalize", &bn_finalize, py::arg("ssum"), py::arg("ssq"),
+          py::arg("R"), py::arg("rmean"), py::arg("rvar"), py::arg("eps"),
+          py::arg("momentum"), py::arg("zero_src") = false);
   mod.def("bn_bwd_dx_only", &bn_bwd_dx_only);
   mod.def("maxpool_fwd", &maxpool_fwd);
   mod.def("gemm_nt_bf16", &gemm_nt_bf16, py::arg("A"), py::arg("Bt"),
@@ -613,11 +645,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gemm_nt_fused", &gemm_nt_fused, py::arg("A"), py::arg("Bt"),
           py::arg("a_scale"), py::arg("a_shift"), py::arg("a_relu"),
           py::arg("want_stats"), py::arg("want_aout") = false,
-          py::arg("c_add") = c10::nullopt);
+          py::arg("c_add") = c10::nullopt,
+          py::arg("ws_sum") = c10::nullopt,
+          py::arg("ws_ssq") = c10::nullopt);
   mod.def("conv3x3_fwd", &conv3x3_fwd);
   mod.def("conv1x1_strided", &conv1x1_strided);
   mod.def("gemm_nt_dgrad_epi", &gemm_nt_dgrad_epi, py::arg("A"),
           py::arg("Bt"), py::arg("y"), py::arg("z"), py::arg("mean"),
-          py::arg("rstd"), py::arg("c_add") = c10::nullopt);
+          py::arg("rstd"), py::arg("c_add") = c10::nullopt,
+          py::arg("ws_db") = c10::nullopt, py::arg("ws_dg") = c10::nullopt);
   mod.def("maxpool_bwd", &maxpool_bwd);
 }

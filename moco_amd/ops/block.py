@@ -43,6 +43,23 @@ import torch.nn.functional as F
 from . import ext, force_fallback, has_ext
 
 
+_WS = {}  # (device, kind, N) -> pre-zeroed [256, N] fp32 workspace
+
+
+def _ws_pair(device, kind, n):
+    """Pre-zeroed slot-reduction workspaces, shared per (device, N): the
+    consuming finalize/slot_reduce kernels re-zero them after reading, so
+    no per-call fill kernels run.  Safe because every producer/consumer
+    pair is stream-ordered on the compute stream."""
+    key = (device, kind, n)
+    ws = _WS.get(key)
+    if ws is None:
+        ws = (torch.zeros(256, n, dtype=torch.float32, device=device),
+              torch.zeros(256, n, dtype=torch.float32, device=device))
+        _WS[key] = ws
+    return ws
+
+
 def _as2d(t4d: torch.Tensor) -> torch.Tensor:
     """channels_last [N,C,H,W] -> the underlying dense [N*H*W, C] buffer
     (stride-free view)."""
@@ -118,8 +135,12 @@ class _FusedBottleneck(torch.autograd.Function):
         _C = ext()
 
         # conv1 (1x1/s1) with bn1-stats epilogue
-        z1, s1, q1 = _C.gemm_nt_fused(x2d, w1b, None, None, False, True)
-        mean1, rstd1 = _C.bn_finalize(s1, q1, M, rm1, rv1, eps[0], mom)
+        Co1 = w1b.shape[0]
+        ws1 = _ws_pair(x.device, "stats", Co1)
+        z1, s1, q1 = _C.gemm_nt_fused(x2d, w1b, None, None, False, True,
+                                      False, None, ws1[0], ws1[1])
+        mean1, rstd1 = _C.bn_finalize(s1, q1, M, rm1, rv1, eps[0], mom,
+                                      zero_src=True)
         a1 = _C.bn_fwd_eval(z1, None, gamma1, beta1, mean1, rstd1, True)
         a1_4d = _as4d(a1, N, H, W)
 
@@ -137,14 +158,17 @@ class _FusedBottleneck(torch.autograd.Function):
 
         # conv3 (1x1/s1): bn2-apply prologue + bn3-stats epilogue
         # (+ a2 materialized when the backward will need it for wgrad)
+        Co3 = w3b.shape[0]
+        ws3 = _ws_pair(x.device, "stats", Co3)
         outs = _C.gemm_nt_fused(z2, w3b, scale2, shift2, True, True,
-                                need_grad, None)
+                                need_grad, None, ws3[0], ws3[1])
         if need_grad:
             z3, s3, q3, a2 = outs
         else:
             z3, s3, q3 = outs
             a2 = None
-        mean3, rstd3 = _C.bn_finalize(s3, q3, M2, rm3, rv3, eps[2], mom)
+        mean3, rstd3 = _C.bn_finalize(s3, q3, M2, rm3, rv3, eps[2], mom,
+                                      zero_src=True)
 
         ident2d = _as2d(identity if identity is not None else x)
         y = _C.bn_fwd_eval(z3, ident2d, gamma3, beta3, mean3, rstd3, True)
@@ -200,8 +224,10 @@ class _FusedBottleneck(torch.autograd.Function):
         # conv3 dgrad + fused bn2 backward-reduce (mask from a2, zhat
         # from z2) — gp2 is the relu-masked da2
         w3t = w3b.t().contiguous()
+        wsr = _ws_pair(z2.device, "dgrad", z2.shape[1])
         gp2, dbeta2, dgamma2 = _C.gemm_nt_dgrad_epi(dz3, w3t, a2, z2,
-                                                    mean2, rstd2)
+                                                    mean2, rstd2, None,
+                                                    wsr[0], wsr[1])
         dz2 = _C.bn_bwd_dx_only(gp2, z2, mean2, rstd2, gamma2, dbeta2,
                                 dgamma2)
         # conv3 wgrad over the materialized a2 (MIOpen wrw igemm: measured
@@ -230,8 +256,9 @@ class _FusedBottleneck(torch.autograd.Function):
         w1t = w1b.t().contiguous()
         x2d = _as2d(x)
         if has_prev and not has_sep_identity and ctx.prev_hand is not None:
+            wsp = _ws_pair(x.device, "dgrad", x2d.shape[1])
             dx2d, db3p, dg3p = _C.gemm_nt_dgrad_epi(
-                dz1, w1t, x2d, z3p, mean3p, rstd3p, dres)
+                dz1, w1t, x2d, z3p, mean3p, rstd3p, dres, wsp[0], wsp[1])
             ctx.prev_hand["deposit"] = (db3p, dg3p)
         else:
             dx2d = _C.gemm_nt_bf16(dz1, w1t,

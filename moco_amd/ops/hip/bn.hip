@@ -93,6 +93,11 @@ __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
 }
 
 // ---- stats pass 2: finalize mean/rstd, update running stats ---------------
+// ZERO_SRC: write zeros back to the partial slabs after reading — the
+// conv-GEMM stats path reuses one pre-zeroed workspace per shape, so the
+// finalize restores the invariant for the next dispatch (no separate
+// fill kernels).
+template <bool ZERO_SRC>
 __global__ __launch_bounds__(256) void bn_stats_final_kernel(
     const float* __restrict__ part_s, const float* __restrict__ part_q,
     float* __restrict__ mean, float* __restrict__ rstd,
@@ -107,6 +112,10 @@ __global__ __launch_bounds__(256) void bn_stats_final_kernel(
     for (int p = pg; p < P; p += 32) {
       s += part_s[(int64_t)p * C + c];
       q += part_q[(int64_t)p * C + c];
+      if (ZERO_SRC) {
+        const_cast<float*>(part_s)[(int64_t)p * C + c] = 0.f;
+        const_cast<float*>(part_q)[(int64_t)p * C + c] = 0.f;
+      }
     }
   __shared__ float lss[256], lqq[256];
   lss[threadIdx.x] = s;
@@ -303,7 +312,7 @@ extern "C" int launch_bn_stats(void* stream, const void* x, float* part_s,
   bn_stats_partial_kernel<<<P, 256, 0, s>>>(
       (const unsigned short*)x, part_s, part_q, R, (int)C);
   HIP_CHECK_LAST();
-  bn_stats_final_kernel<<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
+  bn_stats_final_kernel<false><<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
       part_s, part_q, mean, rstd, running_mean, running_var, R, (int)C,
       P, eps, momentum);
   HIP_CHECK_LAST();
@@ -318,11 +327,56 @@ extern "C" int launch_bn_finalize(void* stream, const float* ssum,
                                   const float* ssq, float* mean, float* rstd,
                                   float* running_mean, float* running_var,
                                   int64_t R, int64_t C, int64_t P, float eps,
-                                  float momentum) {
+                                  float momentum, bool zero_src) {
   hipStream_t s = (hipStream_t)stream;
-  bn_stats_final_kernel<<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
-      ssum, ssq, mean, rstd, running_mean, running_var, R, (int)C, (int)P,
-      eps, momentum);
+  if (zero_src)
+    bn_stats_final_kernel<true>
+        <<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
+            ssum, ssq, mean, rstd, running_mean, running_var, R, (int)C,
+            (int)P, eps, momentum);
+  else
+    bn_stats_final_kernel<false>
+        <<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
+            ssum, ssq, mean, rstd, running_mean, running_var, R, (int)C,
+            (int)P, eps, momentum);
+  HIP_CHECK_LAST();
+  return 0;
+}
+
+// reduce the dgrad-epilogue's two [P, N] slot slabs to [N] totals and
+// re-zero them (same workspace-reuse contract as bn_finalize)
+__global__ __launch_bounds__(256) void slot_reduce2_kernel(
+    float* __restrict__ s1, float* __restrict__ s2, float* __restrict__ o1,
+    float* __restrict__ o2, int C, int P) {
+  const int c = blockIdx.x * 8 + (threadIdx.x & 7);
+  const int pg = threadIdx.x >> 3;
+  float a = 0.f, b = 0.f;
+  if (c < C)
+    for (int p = pg; p < P; p += 32) {
+      a += s1[(int64_t)p * C + c];
+      b += s2[(int64_t)p * C + c];
+      s1[(int64_t)p * C + c] = 0.f;
+      s2[(int64_t)p * C + c] = 0.f;
+    }
+  __shared__ float lsa[256], lsb[256];
+  lsa[threadIdx.x] = a;
+  lsb[threadIdx.x] = b;
+  __syncthreads();
+  if (pg != 0 || c >= C) return;
+  for (int g = 1; g < 32; ++g) {
+    a += lsa[(threadIdx.x & 7) + g * 8];
+    b += lsb[(threadIdx.x & 7) + g * 8];
+  }
+  o1[c] = a;
+  o2[c] = b;
+}
+
+extern "C" int launch_slot_reduce2(void* stream, float* s1, float* s2,
+                                   float* o1, float* o2, int64_t C,
+                                   int64_t P) {
+  hipStream_t s = (hipStream_t)stream;
+  slot_reduce2_kernel<<<dim3((unsigned)((C + 7) / 8)), 256, 0, s>>>(
+      s1, s2, o1, o2, (int)C, (int)P);
   HIP_CHECK_LAST();
   return 0;
 }

@@ -80,7 +80,10 @@ int launch_bn_bwd_dx(void* stream, const void* g, const void* y,
 int launch_bn_finalize(void* stream, const float* ssum, const float* ssq,
                        float* mean, float* rstd, float* running_mean,
                        float* running_var, int64_t R, int64_t C, int64_t P,
-                       float eps, float momentum);
+                       float eps, float momentum, bool zero_src);
+// reduce two [P, C] slot slabs to [C] totals, re-zeroing the slabs
+int launch_slot_reduce2(void* stream, float* s1, float* s2, float* o1,
+                        float* o2, int64_t C, int64_t P);
 
 // ---- augment.hip ----------------------------------------------------------
 // src: [B,3,H,W] fp32; params: [M,12] fp32; out: [M,3,S,S] (bf16 or fp32,
