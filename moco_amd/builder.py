@@ -168,6 +168,12 @@ class MoCo(nn.Module):
                                                      self.shuffle_seed)
             # EMA before the key forward, once per step (reference invariant)
             self._momentum_update_key_encoder()
+            # batched bf16 weight-mirror refresh (ops/mirrors.py): query
+            # masters changed in the last optimizer step, key masters in
+            # the EMA above
+            from .ops.mirrors import refresh_mirrors
+            refresh_mirrors(self.encoder_q)
+            refresh_mirrors(self.encoder_k)
 
         q = self.encoder_q(im_q)
         q = l2_normalize(q, dim=1)

@@ -162,7 +162,26 @@ class ResNet(nn.Module):
                                 norm_layer=norm_layer, **kw))
         return nn.Sequential(*layers)
 
+    def _update_bn_counters(self):
+        """One foreach add for every BN's num_batches_tracked instead of
+        ~53 single-element kernels per forward (bn_act/fused_bottleneck
+        skip their own increments for counters tagged here).  Rebuilt
+        lazily because .to()/.cuda() replace buffer tensors."""
+        cs = getattr(self, "_bn_counters", None)
+        if cs is None or cs[0] is not self.bn1.num_batches_tracked:
+            cs = []
+            for m in self.modules():
+                if isinstance(m, nn.modules.batchnorm._BatchNorm) \
+                        and m.num_batches_tracked is not None:
+                    cs.append(m.num_batches_tracked)
+                    m._nbt_managed = True
+            self._bn_counters = cs
+        if cs:
+            torch._foreach_add_(cs, 1)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.training:
+            self._update_bn_counters()
         x = bn_act(self.conv1(x), self.bn1, relu=True)
         x = stem_maxpool(x)
         x = self.layer1(x)

@@ -191,6 +191,31 @@ void multi_tensor_ema(torch::Tensor pd, torch::Tensor ps, torch::Tensor sizes,
       sizes.data_ptr<int32_t>(), pd.size(0), (float)m));
 }
 
+void multi_cast_bf16(torch::Tensor src_ptrs, torch::Tensor dst_ptrs,
+                     torch::Tensor sizes) {
+  CHECK_CUDA_CONTIG(src_ptrs);
+  CHECK_CUDA_CONTIG(dst_ptrs);
+  CHECK_CUDA_CONTIG(sizes);
+  at::cuda::CUDAGuard guard(src_ptrs.device());
+  CHECK_LAUNCH(launch_multi_cast_bf16(
+      cur_stream(), src_ptrs.data_ptr<int64_t>(),
+      dst_ptrs.data_ptr<int64_t>(), sizes.data_ptr<int32_t>(),
+      src_ptrs.size(0)));
+}
+
+void multi_transpose_bf16(torch::Tensor src_ptrs, torch::Tensor dst_ptrs,
+                          torch::Tensor meta) {
+  CHECK_CUDA_CONTIG(src_ptrs);
+  CHECK_CUDA_CONTIG(dst_ptrs);
+  CHECK_CUDA_CONTIG(meta);
+  TORCH_CHECK(meta.numel() == src_ptrs.size(0) * 4);
+  at::cuda::CUDAGuard guard(src_ptrs.device());
+  CHECK_LAUNCH(launch_multi_transpose_bf16(
+      cur_stream(), src_ptrs.data_ptr<int64_t>(),
+      dst_ptrs.data_ptr<int64_t>(), meta.data_ptr<int32_t>(),
+      src_ptrs.size(0)));
+}
+
 void multi_tensor_sgd(torch::Tensor pp, torch::Tensor pg, torch::Tensor pb,
                       torch::Tensor sizes, double lr, double momentum,
                       double wd) {
@@ -628,6 +653,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("queue_enqueue", &queue_enqueue);
   mod.def("queue_enqueue_dev", &queue_enqueue_dev);
   mod.def("multi_tensor_ema", &multi_tensor_ema);
+  mod.def("multi_cast_bf16", &multi_cast_bf16);
+  mod.def("multi_transpose_bf16", &multi_transpose_bf16);
   mod.def("multi_tensor_sgd", &multi_tensor_sgd);
   mod.def("ema_flat", &ema_flat);
   mod.def("two_crop_augment", &two_crop_augment);

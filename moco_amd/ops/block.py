@@ -128,10 +128,20 @@ class _FusedBottleneck(torch.autograd.Function):
         M = N * H * W
 
         x2d = _as2d(x)
-        w1b = _bf16_weight(w1)
-        w2b = w2.detach().to(torch.bfloat16) \
-            .contiguous(memory_format=torch.channels_last)
-        w3b = _bf16_weight(w3)
+        # bf16 mirrors refreshed once per step by ops/mirrors.py; inline
+        # casts only for standalone-block use (tests)
+        w1b = getattr(w1, "_mirror_bf16", None)
+        if w1b is None:
+            w1b = _bf16_weight(w1)
+        w2b = getattr(w2, "_mirror_bf16", None)
+        if w2b is None:
+            w2b = w2.detach().to(torch.bfloat16) \
+                .contiguous(memory_format=torch.channels_last)
+        w3b = getattr(w3, "_mirror_bf16", None)
+        if w3b is None:
+            w3b = _bf16_weight(w3)
+        w1t = getattr(w1, "_mirror_bf16_t", None)
+        w3t = getattr(w3, "_mirror_bf16_t", None)
         _C = ext()
 
         # conv1 (1x1/s1) with bn1-stats epilogue
@@ -181,10 +191,13 @@ class _FusedBottleneck(torch.autograd.Function):
         ctx.save_for_backward(x, z1, a1, z2, a2 if a2 is not None else z2,
                               z3, y, w1b, w2b, w3b, gamma1, gamma2, gamma3,
                               mean1, rstd1, mean2, rstd2, mean3, rstd3,
-                              scale2, shift2, z3p, mean3p, rstd3p)
+                              scale2, shift2, z3p, mean3p, rstd3p,
+                              w1t if w1t is not None else w1b,
+                              w3t if w3t is not None else w3b)
         ctx.meta = (N, H, W, Ho, Wo, stride, groups,
                     identity is not None, w1.stride(), w3.stride(),
-                    prev_bn3 is not None)
+                    prev_bn3 is not None, w1t is not None,
+                    w3t is not None)
         ctx.prev_hand = prev_hand
         ctx.my_hand = my_hand
         if state_sink is not None:
@@ -198,9 +211,13 @@ class _FusedBottleneck(torch.autograd.Function):
     def backward(ctx, gy):
         (x, z1, a1, z2, a2, z3, y, w1b, w2b, w3b, gamma1, gamma2, gamma3,
          mean1, rstd1, mean2, rstd2, mean3, rstd3, scale2, shift2,
-         z3p, mean3p, rstd3p) = ctx.saved_tensors
+         z3p, mean3p, rstd3p, w1t, w3t) = ctx.saved_tensors
         (N, H, W, Ho, Wo, stride, groups, has_sep_identity, w1_strides,
-         w3_strides, has_prev) = ctx.meta
+         w3_strides, has_prev, has_w1t, has_w3t) = ctx.meta
+        if not has_w1t:
+            w1t = None
+        if not has_w3t:
+            w3t = None
         _C = ext()
 
         gy2d = _as2d(gy.contiguous(memory_format=torch.channels_last))
@@ -223,7 +240,8 @@ class _FusedBottleneck(torch.autograd.Function):
 
         # conv3 dgrad + fused bn2 backward-reduce (mask from a2, zhat
         # from z2) — gp2 is the relu-masked da2
-        w3t = w3b.t().contiguous()
+        if w3t is None:
+            w3t = w3b.t().contiguous()
         wsr = _ws_pair(z2.device, "dgrad", z2.shape[1])
         gp2, dbeta2, dgamma2 = _C.gemm_nt_dgrad_epi(dz3, w3t, a2, z2,
                                                     mean2, rstd2, None,
@@ -253,7 +271,8 @@ class _FusedBottleneck(torch.autograd.Function):
         # (from x = that block's output) and reduces its dbeta3/dgamma3
         # (deposited into prev_hand), so the grad handed upstream is
         # already masked and that block skips its whole reduce pass.
-        w1t = w1b.t().contiguous()
+        if w1t is None:
+            w1t = w1b.t().contiguous()
         x2d = _as2d(x)
         if has_prev and not has_sep_identity and ctx.prev_hand is not None:
             wsp = _ws_pair(x.device, "dgrad", x2d.shape[1])
@@ -311,7 +330,8 @@ def fused_bottleneck(x: torch.Tensor, block) -> Optional[torch.Tensor]:
     else:
         identity = None
     for bn in (block.bn1, block.bn2, block.bn3):
-        if bn.num_batches_tracked is not None:
+        if bn.num_batches_tracked is not None \
+                and not getattr(bn, "_nbt_managed", False):
             bn.num_batches_tracked += 1
     bn_state = (block.bn1.running_mean, block.bn1.running_var,
                 block.bn2.running_mean, block.bn2.running_var,

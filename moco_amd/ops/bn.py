@@ -55,7 +55,8 @@ def bn_act(x: torch.Tensor, bn: torch.nn.BatchNorm2d, relu: bool = True,
     """y = relu?(bn(x) + residual?) with nn.BatchNorm2d semantics
     (biased batch var for normalization, unbiased for the running stats)."""
     training = bn.training
-    if training and bn.num_batches_tracked is not None:
+    if training and bn.num_batches_tracked is not None \
+            and not getattr(bn, "_nbt_managed", False):
         bn.num_batches_tracked += 1
 
     use_fused = (x.is_cuda and x.dtype == torch.bfloat16

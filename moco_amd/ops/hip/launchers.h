@@ -163,3 +163,17 @@ int launch_gemm_nt_dgrad_epi(void* stream, const void* A, const void* Bt,
                              const void* y, const void* z, const float* mean,
                              const float* rstd, float* dbeta, float* dgamma,
                              const void* c_add);
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+// ---- weight mirrors (multi_tensor.hip) ------------------------------------
+int launch_multi_cast_bf16(void* stream, const int64_t* src_ptrs,
+                           const int64_t* dst_ptrs, const int32_t* sizes,
+                           int64_t nchunks);
+int launch_multi_transpose_bf16(void* stream, const int64_t* src_ptrs,
+                                const int64_t* dst_ptrs, const int32_t* meta,
+                                int64_t ntiles);
+#ifdef __cplusplus
+}
+#endif

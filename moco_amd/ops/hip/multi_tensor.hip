@@ -109,3 +109,84 @@ extern "C" int launch_ema_flat(void* stream, float* dst, const float* src,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---- weight mirrors --------------------------------------------------------
+// One launch refreshes every conv weight's bf16 working copy from the fp32
+// master (the fused-block GEMMs read bf16 mirrors; per-weight .to(bfloat16)
+// calls were ~115 launches/step).  Chunk table like the EMA/SGD kernels.
+namespace {
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+__global__ __launch_bounds__(256) void multi_cast_bf16_kernel(
+    const int64_t* __restrict__ src_ptrs, const int64_t* __restrict__ dst_ptrs,
+    const int32_t* __restrict__ sizes) {
+  const float* __restrict__ src =
+      reinterpret_cast<const float*>(src_ptrs[blockIdx.x]);
+  bf16* __restrict__ dst = reinterpret_cast<bf16*>(dst_ptrs[blockIdx.x]);
+  const int n = sizes[blockIdx.x];
+  for (int i = threadIdx.x * 4; i < n; i += 256 * 4) {
+    if (i + 4 <= n) {
+      const float4 v = *reinterpret_cast<const float4*>(src + i);
+      bf16x4 o = {f2bf(v.x), f2bf(v.y), f2bf(v.z), f2bf(v.w)};
+      *reinterpret_cast<bf16x4*>(dst + i) = o;
+    } else {
+      for (int j = i; j < n; ++j) dst[j] = f2bf(src[j]);
+    }
+  }
+}
+
+// One launch transposes every 1x1-conv bf16 mirror [Co, Ci] -> [Ci, Co]
+// (the dgrad GEMMs want the contraction contiguous).  One workgroup per
+// 64x64 tile; meta = {co0, ci0, Co, Ci} per tile; Co, Ci % 64 == 0.
+__global__ __launch_bounds__(256) void multi_transpose_bf16_kernel(
+    const int64_t* __restrict__ src_ptrs, const int64_t* __restrict__ dst_ptrs,
+    const int32_t* __restrict__ meta) {
+  __shared__ bf16 tile[64][72];  // 72: pad away bank conflicts
+  const bf16* __restrict__ src =
+      reinterpret_cast<const bf16*>(src_ptrs[blockIdx.x]);
+  bf16* __restrict__ dst = reinterpret_cast<bf16*>(dst_ptrs[blockIdx.x]);
+  const int co0 = meta[blockIdx.x * 4 + 0];
+  const int ci0 = meta[blockIdx.x * 4 + 1];
+  const int Co = meta[blockIdx.x * 4 + 2];
+  const int Ci = meta[blockIdx.x * 4 + 3];
+  // load 64 rows x 64 cols: thread t covers row t/4, cols (t%4)*16..+16
+  const int lr = threadIdx.x / 4;
+  const int lc = (threadIdx.x % 4) * 16;
+#pragma unroll
+  for (int j = 0; j < 16; j += 8)
+    *reinterpret_cast<bf16x8*>(&tile[lr][lc + j]) =
+        *reinterpret_cast<const bf16x8*>(src + (int64_t)(co0 + lr) * Ci +
+                                         ci0 + lc + j);
+  __syncthreads();
+  bf16x8 o[2];
+#pragma unroll
+  for (int p = 0; p < 2; ++p)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[p][j] = tile[lc + p * 8 + j][lr];
+#pragma unroll
+  for (int p = 0; p < 2; ++p)
+    *reinterpret_cast<bf16x8*>(dst + (int64_t)(ci0 + lr) * Co + co0 + lc +
+                               p * 8) = o[p];
+}
+}  // namespace
+
+extern "C" int launch_multi_cast_bf16(void* stream, const int64_t* src_ptrs,
+                                      const int64_t* dst_ptrs,
+                                      const int32_t* sizes, int64_t nchunks) {
+  hipStream_t s = (hipStream_t)stream;
+  multi_cast_bf16_kernel<<<dim3((unsigned)nchunks), 256, 0, s>>>(
+      src_ptrs, dst_ptrs, sizes);
+  HIP_CHECK_LAST();
+  return 0;
+}
+
+extern "C" int launch_multi_transpose_bf16(void* stream,
+                                           const int64_t* src_ptrs,
+                                           const int64_t* dst_ptrs,
+                                           const int32_t* meta,
+                                           int64_t ntiles) {
+  hipStream_t s = (hipStream_t)stream;
+  multi_transpose_bf16_kernel<<<dim3((unsigned)ntiles), 256, 0, s>>>(
+      src_ptrs, dst_ptrs, meta);
+  HIP_CHECK_LAST();
+  return 0;
+}
