@@ -41,6 +41,8 @@ def get_args():
     p.add_argument("--pool-batches", type=int, default=4)
     p.add_argument("--phase-timing", action="store_true",
                    help="print per-phase GPU-event ms after the timed run")
+    p.add_argument("--print-mem", action="store_true",
+                   help="print peak device memory after the timed run")
     return p.parse_args()
 
 
@@ -133,6 +135,12 @@ def main():
     total_images = args.batch_per_gpu * n_gpus * args.steps
     img_s = total_images / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
+
+    if args.print_mem and not cpu_fallback and rank == 0:
+        print("peak_mem_gb:",
+              round(torch.cuda.max_memory_allocated() / 2**30, 2),
+              "reserved_gb:",
+              round(torch.cuda.max_memory_reserved() / 2**30, 2))
 
     if rank == 0:
         result = {
