@@ -60,10 +60,17 @@ class MoCo(nn.Module):
             param_k.requires_grad = False
 
         # canonical queue: [dim, K] fp32, L2-normalized columns
-        # (reference builder.py:36-38)
+        # (reference builder.py:36-38).  Drawn from a dedicated generator
+        # seeded by shuffle_seed so the ring is replicated across ranks BY
+        # CONSTRUCTION even when each rank seeds its global RNG differently
+        # (DDP syncs parameters at wrap time but we run
+        # broadcast_buffers=False, so nothing else would ever reconcile a
+        # divergent initial queue).
+        qgen = torch.Generator().manual_seed(0x5EED ^ shuffle_seed)
         self.register_buffer("queue",
-                             nn.functional.normalize(torch.randn(dim, K),
-                                                     dim=0))
+                             nn.functional.normalize(
+                                 torch.randn(dim, K, generator=qgen),
+                                 dim=0))
         self.register_buffer("queue_ptr", torch.zeros(1, dtype=torch.long))
 
         self._ema: Optional[MultiTensorEMA] = None

@@ -141,3 +141,51 @@ def test_shuffle_exchange_world4():
 
 def test_queue_replicated_world4():
     _run_spawn(_worker_queue_replicated, 29317, world=4)
+
+
+def _worker_bench_codepath(rank, world, port):
+    """bench.py's exact training-step code path (MoCoStepper + DDP +
+    FusedSGD) at gloo world 2: after K steps, params and the queue must be
+    bit-identical across ranks (DDP grad sync + deterministic enqueue)."""
+    _init(rank, world, port)
+    import moco_amd
+    from moco_amd.engine import MoCoStepper
+    from moco_amd.models.resnet import build_model
+    from moco_amd.ops import FusedSGD
+
+    torch.manual_seed(1234 + rank)  # bench.py seeds per rank
+    model = moco_amd.MoCo(
+        lambda num_classes: build_model("resnet18", num_classes=num_classes),
+        dim=128, K=64, m=0.999, T=0.2, mlp=True, shuffle_seed=1234)
+    model = torch.nn.parallel.DistributedDataParallel(
+        model, broadcast_buffers=False)
+    optimizer = FusedSGD(model.parameters(), lr=0.03, momentum=0.9,
+                         weight_decay=1e-4)
+    stepper = MoCoStepper(model, optimizer, torch.device("cpu"),
+                          aug_plus=True, out_size=32,
+                          dtype=torch.float32)
+    g = torch.Generator().manual_seed(4321 + rank)
+    pool = [torch.rand(4, 3, 40, 40, generator=g) for _ in range(2)]
+    losses = []
+    for i in range(4):
+        loss, _, _ = stepper.step_from_raw(pool[i % 2])
+        losses.append(float(loss))
+    assert all(l == l and l < 100 for l in losses), losses
+
+    raw = model.module
+    # queue replicated bit-exactly across ranks
+    qs = [torch.empty_like(raw.queue) for _ in range(world)]
+    dist.all_gather(qs, raw.queue)
+    assert torch.equal(qs[0], qs[1]), "queue diverged across ranks"
+    assert raw.queue_ptr.item() == (4 * 4 * world) % 64
+    # params identical across ranks after DDP-synced updates
+    flat = torch.cat([p.detach().reshape(-1)
+                      for p in raw.parameters()])[:100000]
+    ps = [torch.empty_like(flat) for _ in range(world)]
+    dist.all_gather(ps, flat)
+    assert torch.equal(ps[0], ps[1]), "params diverged across ranks"
+    dist.destroy_process_group()
+
+
+def test_bench_codepath_world2_numerics():
+    _run_spawn(_worker_bench_codepath, 29318)
