@@ -235,6 +235,7 @@ def train(loader, model, criterion, optimizer, epoch, device, args):
 
 @torch.no_grad()
 def validate(loader, model, criterion, device, args):
+    losses = AverageMeter("Loss", ":.4e")
     top1 = AverageMeter("Acc@1", ":6.2f")
     top5 = AverageMeter("Acc@5", ":6.2f")
     model.eval()
@@ -245,11 +246,13 @@ def validate(loader, model, criterion, device, args):
         images = torch.stack([tfm(x) for x in raw]).to(device)
         target = target.to(device)
         output = model(images)
-        criterion(output, target)
+        loss = criterion(output, target)
+        losses.update(loss.item(), images.size(0))
         acc1, acc5 = accuracy(output, target, topk=(1, 5))
         top1.update(acc1[0].item(), images.size(0))
         top5.update(acc5[0].item(), images.size(0))
-    print(f" * Acc@1 {top1.avg:.3f} Acc@5 {top5.avg:.3f}")
+    print(f" * Acc@1 {top1.avg:.3f} Acc@5 {top5.avg:.3f} "
+          f"Loss {losses.avg:.4f}")
     return top1.avg
 
 

@@ -401,6 +401,39 @@ std::vector<torch::Tensor> gemm_nt_fused(
   return out;
 }
 
+// fp8 e4m3 GEMM (MX-scaled MFMA, per-tensor software scaling):
+// C_bf16 = (A8 @ B8t^T) * out_scale.  want_stats adds the bn-stats
+// epilogue ([256, N] slot slabs like the bf16 path).
+std::vector<torch::Tensor> gemm_nt_fp8(torch::Tensor A, torch::Tensor Bt,
+                                       double out_scale, bool want_stats,
+                                       c10::optional<torch::Tensor> ws_sum,
+                                       c10::optional<torch::Tensor> ws_ssq) {
+  CHECK_CUDA_CONTIG(A);
+  CHECK_CUDA_CONTIG(Bt);
+  TORCH_CHECK(A.scalar_type() == torch::kFloat8_e4m3fn &&
+              Bt.scalar_type() == torch::kFloat8_e4m3fn &&
+              A.size(1) == Bt.size(1));
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  at::cuda::CUDAGuard guard(A.device());
+  auto C = torch::empty({M, N}, A.options().dtype(torch::kBFloat16));
+  torch::Tensor ssum, ssq;
+  if (want_stats) {
+    if (ws_sum.has_value()) {
+      ssum = *ws_sum;
+      ssq = *ws_ssq;
+    } else {
+      ssum = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+      ssq = torch::zeros({256, N}, A.options().dtype(torch::kFloat32));
+    }
+  }
+  CHECK_LAUNCH(launch_gemm_nt_fp8(
+      cur_stream(), A.data_ptr(), Bt.data_ptr(), C.data_ptr(), M, N, K,
+      (float)out_scale, want_stats ? ssum.data_ptr<float>() : nullptr,
+      want_stats ? ssq.data_ptr<float>() : nullptr));
+  if (want_stats) return {C, ssum, ssq};
+  return {C};
+}
+
 // ---------------------------------------------------------------------------
 // Granular BN entry points for the conv+BN fused block path.
 
@@ -669,6 +702,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("maxpool_fwd", &maxpool_fwd);
   mod.def("gemm_nt_bf16", &gemm_nt_bf16, py::arg("A"), py::arg("Bt"),
           py::arg("c_add") = c10::nullopt);
+  mod.def("gemm_nt_fp8", &gemm_nt_fp8, py::arg("A"), py::arg("Bt"),
+          py::arg("out_scale"), py::arg("want_stats") = false,
+          py::arg("ws_sum") = c10::nullopt, py::arg("ws_ssq") = c10::nullopt);
   mod.def("gemm_nt_fused", &gemm_nt_fused, py::arg("A"), py::arg("Bt"),
           py::arg("a_scale"), py::arg("a_shift"), py::arg("a_relu"),
           py::arg("want_stats"), py::arg("want_aout") = false,

@@ -543,3 +543,173 @@ extern "C" int launch_gemm_nt_dgrad_epi(void* stream, const void* A,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3) GEMM on the MX-scaled MFMA — BASELINE config-4 path.
+// gfx950's NON-scaled fp8 MFMA runs at the bf16 rate; the 2x fp8 rate is
+// only reachable through v_mfma_scale_f32_16x16x128_f8f6f4 (guide §3/§4).
+// Per-TENSOR software scaling: operands are pre-scaled into e4m3 range at
+// quantization, the hardware per-block E8M0 scales are set to 1.0
+// (byte 127), and the epilogue multiplies the fp32 accumulators by
+// sa*sb.  A BK=128-fp8 K-tile is byte-identical to the bf16 BK=64 tile
+// (128 B/row), so the stage_tile/XOR-swizzle machinery is reused as-is.
+// ---------------------------------------------------------------------------
+namespace {
+
+typedef char f8x32 __attribute__((ext_vector_type(32)));
+
+__device__ __forceinline__ f8x32 lds_frag32(const char* smem, int row,
+                                            int kbyte) {
+  union {
+    f8x32 v;
+    bf16x8 h[2];
+  } u;
+  u.h[0] = *reinterpret_cast<const bf16x8*>(smem + row * 128 +
+                                            (kbyte ^ ((row & 7) << 4)));
+  u.h[1] = *reinterpret_cast<const bf16x8*>(
+      smem + row * 128 + ((kbyte + 16) ^ ((row & 7) << 4)));
+  return u.v;
+}
+
+template <int BNT>
+__global__ __launch_bounds__(256, 2) void gemm_nt_fp8_kernel(
+    const char* __restrict__ A, const char* __restrict__ Bt,
+    bf16* __restrict__ C, int64_t M, int N, int K, float out_scale,
+    float* __restrict__ stats_sum, float* __restrict__ stats_sumsq) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int MR = BNT == 128 ? 4 : 2;
+  constexpr int ABYTES = BM * 128;   // 128 rows x 128 B
+  constexpr int BBYTES = BNT * 128;
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int wr = BNT == 128 ? (wave >> 1) : wave;
+  const int wc = BNT == 128 ? (wave & 1) : 0;
+  const int wrows = BNT == 128 ? 64 : 32;
+  const int64_t m0 = (int64_t)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BNT;
+
+  f32x4 acc[MR][4] = {};
+  // stage via the bf16 path: row stride K bytes = K/2 "bf16 columns"
+  stage_tile<BM>((const bf16*)(A + m0 * K), K / 2, smem, wave, lane);
+  stage_tile<BNT>((const bf16*)(Bt + (int64_t)n0 * K), K / 2, smem + ABYTES,
+                  wave, lane);
+  __syncthreads();
+
+  int cur = 0;
+  const int ntiles = K / 128;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      char* nbuf = smem + (cur ^ 1) * (ABYTES + BBYTES);
+      stage_tile<BM>((const bf16*)(A + m0 * K + (t + 1) * 128), K / 2, nbuf,
+                     wave, lane);
+      stage_tile<BNT>((const bf16*)(Bt + (int64_t)n0 * K + (t + 1) * 128),
+                      K / 2, nbuf + ABYTES, wave, lane);
+    }
+    const char* abuf = smem + cur * (ABYTES + BBYTES);
+    const char* bbuf = abuf + ABYTES;
+    const int kbyte = (lane >> 4) * 32;  // lane's 32-byte K-block
+    f8x32 af[MR], bf[4];
+#pragma unroll
+    for (int mr = 0; mr < MR; ++mr)
+      af[mr] = lds_frag32(abuf, wr * wrows + mr * 16 + (lane & 15), kbyte);
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr)
+      bf[nr] = lds_frag32(bbuf, wc * 64 + nr * 16 + (lane & 15), kbyte);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+      for (int nr = 0; nr < 4; ++nr)
+        acc[mr][nr] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[mr], bf[nr], acc[mr][nr], /*cbsz=*/0, /*blgp=*/0,
+            /*opselA=*/0, /*scaleA=*/127, /*opselB=*/0, /*scaleB=*/127);
+    __builtin_amdgcn_s_setprio(0);
+    if (t + 1 < ntiles) __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: per-tensor dequant scale, optional bn-stats accumulation,
+  // widened LDS-transpose store (same structure as the bf16 kernel)
+#pragma unroll
+  for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[mr][nr][r] *= out_scale;
+
+  if (stats_sum != nullptr) {
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      float ssum = 0.f, ssq = 0.f;
+#pragma unroll
+      for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float v = acc[mr][nr][r];
+          ssum += v;
+          ssq += v * v;
+        }
+      ssum += __shfl_xor(ssum, 16, WAVE);
+      ssq += __shfl_xor(ssq, 16, WAVE);
+      ssum += __shfl_xor(ssum, 32, WAVE);
+      ssq += __shfl_xor(ssq, 32, WAVE);
+      if ((lane >> 4) == 0) {
+        const int64_t slot = (int64_t)(blockIdx.x % STAT_SLOTS) * N;
+        const int c = n0 + wc * 64 + nr * 16 + lane;
+        atomicAdd(&stats_sum[slot + c], ssum);
+        atomicAdd(&stats_sumsq[slot + c], ssq);
+      }
+    }
+  }
+
+  __syncthreads();
+  bf16* ct = reinterpret_cast<bf16*>(smem) + wave * wrows * 64;
+#pragma unroll
+  for (int mr = 0; mr < MR; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 4; ++nr) {
+      const int col = nr * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int trow = mr * 16 + (lane >> 4) * 4 + r;
+        ct[trow * 64 + col] = f2bf(acc[mr][nr][r]);
+      }
+    }
+  __syncthreads();
+  const int npieces = wrows * 64 / 8 / WAVE;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    if (i >= npieces) break;
+    const int piece = i * WAVE + lane;
+    const int trow = piece / 8;
+    const int tcol = (piece % 8) * 8;
+    const int64_t row = m0 + wr * wrows + trow;
+    *reinterpret_cast<bf16x8*>(C + row * N + n0 + wc * 64 + tcol) =
+        *reinterpret_cast<const bf16x8*>(ct + trow * 64 + tcol);
+  }
+}
+
+}  // namespace
+
+extern "C" int launch_gemm_nt_fp8(void* stream, const void* A,
+                                  const void* Bt, void* C, int64_t M,
+                                  int64_t N, int64_t K, float out_scale,
+                                  float* stats_sum, float* stats_sumsq) {
+  if (M % 128 != 0 || K % 128 != 0 || N % 64 != 0)
+    return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int BNT = (N % 128 == 0) ? 128 : 64;
+  const bool dbuf = (K / 128) > 1;
+  const int smem = (BM * 128 + BNT * 128) * (dbuf ? 2 : 1);
+  dim3 grid((unsigned)(M / 128), (unsigned)(N / BNT));
+  if (BNT == 128)
+    gemm_nt_fp8_kernel<128><<<grid, 256, smem, s>>>(
+        (const char*)A, (const char*)Bt, (bf16*)C, M, (int)N, (int)K,
+        out_scale, stats_sum, stats_sumsq);
+  else
+    gemm_nt_fp8_kernel<64><<<grid, 256, smem, s>>>(
+        (const char*)A, (const char*)Bt, (bf16*)C, M, (int)N, (int)K,
+        out_scale, stats_sum, stats_sumsq);
+  HIP_CHECK_LAST();
+  return 0;
+}

@@ -177,3 +177,13 @@ int launch_multi_transpose_bf16(void* stream, const int64_t* src_ptrs,
 #ifdef __cplusplus
 }
 #endif
+
+#ifdef __cplusplus
+extern "C"
+#endif
+// fp8 e4m3 GEMM on the MX-scaled MFMA (per-tensor software scaling):
+// C_bf16[M,N] = (A8[M,K] @ B8t[N,K]^T) * out_scale; optional bn-stats
+// epilogue into [256, N] slot slabs.  M%128, N%64, K%128.
+int launch_gemm_nt_fp8(void* stream, const void* A, const void* Bt, void* C,
+                       int64_t M, int64_t N, int64_t K, float out_scale,
+                       float* stats_sum, float* stats_sumsq);

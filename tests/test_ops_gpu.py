@@ -714,3 +714,74 @@ def test_bn_bwd_dx_only_matches_formula():
     ref = gamma[None, :] * rstd[None, :] * (
         gp.float() - dbeta[None, :] / R - zhat * dgamma[None, :] / R)
     assert (dx.float() - ref).abs().max() < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# fp8 (e4m3) GEMM on the MX-scaled MFMA (BASELINE config 4)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("M,N,K", [(256, 64, 128), (1536, 128, 256),
+                                   (128, 192, 512)])
+def test_gemm_nt_fp8_matches_dequant_oracle(M, N, K):
+    """The fp8 GEMM must match the fp32 product of the DEQUANTIZED
+    operands (same values, fp32 accumulate) to bf16-output rounding."""
+    from moco_amd import _C
+    torch.manual_seed(50)
+    A8 = (torch.randn(M, K) * 100).clamp(-448, 448).cuda() \
+        .to(torch.float8_e4m3fn)
+    B8 = (torch.randn(N, K) * 100).clamp(-448, 448).cuda() \
+        .to(torch.float8_e4m3fn)
+    out_scale = 1.0 / (64.0 * 64.0)
+    (C,) = _C.gemm_nt_fp8(A8, B8, out_scale)
+    ref = (A8.float() @ B8.float().t()) * out_scale
+    d = (C.float() - ref).abs()
+    tol = ref.abs().max() * 8e-3 + 1e-2
+    assert d.max() < tol, (d.max(), ref.abs().max())
+
+
+def test_gemm_nt_fp8_identity_layout():
+    """One-hot rows against an asymmetric B catch fragment transposes."""
+    from moco_amd import _C
+    M, N, K = 128, 64, 128
+    A = torch.zeros(M, K)
+    for m in range(M):
+        A[m, (m * 7 + 3) % K] = 2.0
+    B = (torch.arange(N * K, dtype=torch.float32).view(N, K) % 13) - 6
+    A8 = A.cuda().to(torch.float8_e4m3fn)
+    B8 = B.cuda().to(torch.float8_e4m3fn)
+    (C,) = _C.gemm_nt_fp8(A8, B8, 0.5)
+    ref = (A8.float() @ B8.float().t()) * 0.5
+    assert torch.allclose(C.float(), ref, atol=1e-2), \
+        (C.float() - ref).abs().max()
+
+
+def test_conv1x1_fp8_drift_vs_bf16():
+    """Config-4 drift check: fp8 conv fwd vs the bf16 path on realistic
+    (BN-normalized) activations — relative error must stay in the
+    low-percent range expected of per-tensor e4m3."""
+    from moco_amd.ops.fp8 import conv1x1_fp8_fwd
+    torch.manual_seed(51)
+    N, Ci, H, Co = 2, 256, 8, 64
+    x = torch.randn(N, Ci, H, H).cuda()
+    w = (torch.randn(Co, Ci, 1, 1) * 0.05).cuda()
+    ref = torch.nn.functional.conv2d(x, w)
+    out = conv1x1_fp8_fwd(
+        x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last),
+        w)
+    rel = (out.float() - ref).norm() / ref.norm()
+    assert float(rel) < 0.04, float(rel)
+    # and strictly more drift than bf16, but same ballpark (sanity)
+    assert float(rel) > 1e-4
+
+
+def test_gemm_nt_fp8_stats_epilogue():
+    from moco_amd import _C
+    torch.manual_seed(52)
+    M, N, K = 256, 64, 128
+    A8 = (torch.randn(M, K) * 50).cuda().to(torch.float8_e4m3fn)
+    B8 = (torch.randn(N, K) * 50).cuda().to(torch.float8_e4m3fn)
+    s = 1.0 / 2500.0
+    C, ssum, ssq = _C.gemm_nt_fp8(A8, B8, s, True)
+    ref = (A8.float() @ B8.float().t()) * s
+    assert torch.allclose(ssum.sum(0), ref.sum(0), rtol=5e-3, atol=0.5)
+    assert torch.allclose(ssq.sum(0), (ref ** 2).sum(0), rtol=8e-3, atol=1.0)
