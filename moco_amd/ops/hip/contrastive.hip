@@ -32,6 +32,16 @@ __device__ __forceinline__ bf16x8 load_f32_as_bf16x8(const float* p) {
   return r;
 }
 
+// 4-byte-aligned-only variant for the logits-grad rows: their stride is
+// 1 + K (odd), so 16-byte vector loads would be misaligned for most rows
+// (dword-tolerant loads are a gfx950 behavior, not an ISA guarantee)
+__device__ __forceinline__ bf16x8 load_f32_as_bf16x8_u(const float* p) {
+  bf16x8 r;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) r[i] = (__bf16)p[i];
+  return r;
+}
+
 // ---------------------------------------------------------------------------
 // forward: l_neg GEMM.  Block = 4 waves; BM=16 rows x BN=256 cols
 // (wave w covers cols [w*64, w*64+64) as 4 16-col fragments).
@@ -118,7 +128,7 @@ __global__ __launch_bounds__(256) void contrastive_bwd_neg_kernel(
   f32x4 acc = {};
   for (int64_t j0 = jlo; j0 < jhi; j0 += 32) {
     const bf16x8 a =
-        load_f32_as_bf16x8(g + (int64_t)arow * ldg + 1 + j0 + koff);
+        load_f32_as_bf16x8_u(g + (int64_t)arow * ldg + 1 + j0 + koff);
     const bf16x8 b =
         load_f32_as_bf16x8(queue_ck + (int64_t)brow * K + j0 + koff);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);

@@ -22,8 +22,23 @@ __global__ __launch_bounds__(256) void ce0_fwd_kernel(
   // exp(m - nm) = exp(finite), never (-inf) - (-inf) = nan.
   float m = -3.0e38f;
   float s = 0.f;
-  const int64_t M4 = M & ~3LL;
-  for (int64_t i = (int64_t)tid * 4; i < M4; i += 256 * 4) {
+  // peel a misaligned head: the row stride M = 1 + K is odd, so lr is
+  // only 4-byte aligned for row > 0 — 16-byte vector loads must start at
+  // an aligned element (dword-tolerant loads are a gfx950
+  // behavior, not an ISA guarantee)
+  const int head0 = (int)((4 - ((((unsigned long long)(size_t)lr) >> 2)
+                               & 3)) & 3);
+  const int64_t head = head0 < M ? head0 : M;
+  for (int64_t i = tid; i < head; i += 256) {
+    const float v = lr[i];
+    if (v > m) {
+      s *= __expf(m - v);
+      m = v;
+    }
+    s += __expf(v - m);
+  }
+  const int64_t M4 = head + ((M - head) & ~3LL);
+  for (int64_t i = head + (int64_t)tid * 4; i < M4; i += 256 * 4) {
     const float4 v = *reinterpret_cast<const float4*>(lr + i);
     const float vm = fmaxf(fmaxf(v.x, v.y), fmaxf(v.z, v.w));
     if (vm > m) {
