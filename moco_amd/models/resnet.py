@@ -21,7 +21,7 @@ import torch.nn as nn
 from ..ops.block import fused_bottleneck
 from ..ops.bn import bn_act
 from ..ops.conv1x1 import conv1x1 as conv1x1_op
-from ..ops.pool import stem_maxpool
+from ..ops.pool import head_avgpool, stem_maxpool
 
 
 def conv3x3(cin: int, cout: int, stride: int = 1, dilation: int = 1,
@@ -188,7 +188,7 @@ class ResNet(nn.Module):
         x = self.layer2(x)
         x = self.layer3(x)
         x = self.layer4(x)
-        x = self.avgpool(x)
+        x = head_avgpool(x)
         x = torch.flatten(x, 1)
         return self.fc(x)
 

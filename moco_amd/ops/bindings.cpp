@@ -629,6 +629,29 @@ std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x) {
   return {y, idx};
 }
 
+// head adaptive-avg-pool to 1x1 (NHWC bf16)
+torch::Tensor avgpool_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && is_bf16(x) && x.dim() == 4);
+  const int64_t N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  at::cuda::CUDAGuard guard(x.device());
+  auto y = torch::empty({N, C, 1, 1}, x.options())
+               .contiguous(torch::MemoryFormat::ChannelsLast);
+  CHECK_LAUNCH(launch_avgpool_fwd(cur_stream(), x.data_ptr(), y.data_ptr(),
+                                  N, C, H * W));
+  return y;
+}
+
+torch::Tensor avgpool_bwd(torch::Tensor g, int64_t H, int64_t W) {
+  TORCH_CHECK(g.is_cuda() && is_bf16(g));
+  const int64_t N = g.size(0), C = g.size(1);
+  at::cuda::CUDAGuard guard(g.device());
+  auto dx = torch::empty({N, C, H, W}, g.options())
+                .contiguous(torch::MemoryFormat::ChannelsLast);
+  CHECK_LAUNCH(launch_avgpool_bwd(cur_stream(), g.data_ptr(), dx.data_ptr(),
+                                  N, C, H * W));
+  return dx;
+}
+
 torch::Tensor maxpool_bwd(torch::Tensor g, torch::Tensor idx, int64_t H,
                           int64_t W) {
   TORCH_CHECK(g.is_cuda() && is_bf16(g) && g.dim() == 4);
@@ -718,4 +741,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           py::arg("rstd"), py::arg("c_add") = c10::nullopt,
           py::arg("ws_db") = c10::nullopt, py::arg("ws_dg") = c10::nullopt);
   mod.def("maxpool_bwd", &maxpool_bwd);
+  mod.def("avgpool_fwd", &avgpool_fwd);
+  mod.def("avgpool_bwd", &avgpool_bwd);
 }

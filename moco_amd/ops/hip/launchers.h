@@ -187,3 +187,15 @@ extern "C"
 int launch_gemm_nt_fp8(void* stream, const void* A, const void* Bt, void* C,
                        int64_t M, int64_t N, int64_t K, float out_scale,
                        float* stats_sum, float* stats_sumsq);
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+// head adaptive-avg-pool to 1x1, NHWC bf16
+int launch_avgpool_fwd(void* stream, const void* x, void* y, int64_t N,
+                       int64_t C, int64_t HW);
+int launch_avgpool_bwd(void* stream, const void* g, void* dx, int64_t N,
+                       int64_t C, int64_t HW);
+#ifdef __cplusplus
+}
+#endif

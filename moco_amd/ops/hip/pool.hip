@@ -158,3 +158,84 @@ extern "C" int launch_maxpool_bwd(void* stream, const void* g,
   HIP_CHECK_LAST();
   return 0;
 }
+
+// ---------------------------------------------------------------------------
+// Head adaptive-avg-pool to 1x1 (the ResNet global pool), NHWC bf16.
+// fwd: one block per image; each thread averages its channel octets over
+// the HW positions.  bwd: broadcast g/HW (elementwise).
+// ---------------------------------------------------------------------------
+namespace {
+
+__device__ __forceinline__ unsigned short f2us_(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  const unsigned int lsb = (v.i >> 16) & 1;
+  v.i += 0x7fffu + lsb;
+  return (unsigned short)(v.i >> 16);
+}
+
+__global__ __launch_bounds__(256) void avgpool_fwd_kernel(
+    const unsigned short* __restrict__ x, unsigned short* __restrict__ y,
+    int C, int HW) {
+  const int n = blockIdx.x;
+  const int Q = C / 8;
+  const float inv = 1.0f / (float)HW;
+  for (int q = threadIdx.x; q < Q; q += 256) {
+    float acc[8] = {};
+    const unsigned short* base = x + ((int64_t)n * HW) * C + q * 8;
+    for (int p = 0; p < HW; ++p) {
+      const ushort8v v =
+          *reinterpret_cast<const ushort8v*>(base + (int64_t)p * C);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += us2f_(v[j]);
+    }
+    ushort8v out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2us_(acc[j] * inv);
+    *reinterpret_cast<ushort8v*>(y + (int64_t)n * C + q * 8) = out;
+  }
+}
+
+__global__ __launch_bounds__(256) void avgpool_bwd_kernel(
+    const unsigned short* __restrict__ g, unsigned short* __restrict__ dx,
+    int C, int HW, int64_t total8) {
+  const float inv = 1.0f / (float)HW;
+  const int Q = C / 8;
+  for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < total8;
+       i += (int64_t)gridDim.x * 256) {
+    const int64_t q = i % Q;          // channel octet
+    const int64_t n = i / (Q * (int64_t)HW);
+    const ushort8v gv =
+        *reinterpret_cast<const ushort8v*>(g + (n * Q + q) * 8);
+    ushort8v out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2us_(us2f_(gv[j]) * inv);
+    *reinterpret_cast<ushort8v*>(dx + i * 8) = out;
+  }
+}
+
+}  // namespace
+
+extern "C" int launch_avgpool_fwd(void* stream, const void* x, void* y,
+                                  int64_t N, int64_t C, int64_t HW) {
+  if (C % 8 != 0) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  avgpool_fwd_kernel<<<dim3((unsigned)N), 256, 0, s>>>(
+      (const unsigned short*)x, (unsigned short*)y, (int)C, (int)HW);
+  HIP_CHECK_LAST();
+  return 0;
+}
+
+extern "C" int launch_avgpool_bwd(void* stream, const void* g, void* dx,
+                                  int64_t N, int64_t C, int64_t HW) {
+  if (C % 8 != 0) return (int)hipErrorInvalidValue;
+  hipStream_t s = (hipStream_t)stream;
+  const int64_t total8 = N * HW * C / 8;
+  int64_t b = (total8 + 255) / 256;
+  if (b > 2048) b = 2048;
+  avgpool_bwd_kernel<<<dim3((unsigned)b), 256, 0, s>>>(
+      (const unsigned short*)g, (unsigned short*)dx, (int)C, (int)HW,
+      total8);
+  HIP_CHECK_LAST();
+  return 0;
+}

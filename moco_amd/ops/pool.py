@@ -31,3 +31,25 @@ def stem_maxpool(x: torch.Tensor) -> torch.Tensor:
             and x.is_contiguous(memory_format=torch.channels_last)):
         return _MaxPool3x3s2.apply(x)
     return F.max_pool2d(x, kernel_size=3, stride=2, padding=1)
+
+
+class _HeadAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_hw = (x.shape[2], x.shape[3])
+        return ext().avgpool_fwd(x)
+
+    @staticmethod
+    def backward(ctx, g):
+        H, W = ctx.in_hw
+        return ext().avgpool_bwd(
+            g.contiguous(memory_format=torch.channels_last), H, W)
+
+
+def head_avgpool(x: torch.Tensor) -> torch.Tensor:
+    """AdaptiveAvgPool2d((1, 1)) — the ResNet global pool."""
+    if (x.is_cuda and x.dtype == torch.bfloat16 and not force_fallback()
+            and has_ext() and x.shape[1] % 8 == 0
+            and x.is_contiguous(memory_format=torch.channels_last)):
+        return _HeadAvgPool.apply(x)
+    return F.adaptive_avg_pool2d(x, (1, 1))

@@ -785,3 +785,20 @@ def test_gemm_nt_fp8_stats_epilogue():
     ref = (A8.float() @ B8.float().t()) * s
     assert torch.allclose(ssum.sum(0), ref.sum(0), rtol=5e-3, atol=0.5)
     assert torch.allclose(ssq.sum(0), (ref ** 2).sum(0), rtol=8e-3, atol=1.0)
+
+
+def test_head_avgpool_fwd_bwd():
+    from moco_amd.ops.pool import head_avgpool
+    torch.manual_seed(60)
+    for N, C, H in [(2, 2048, 7), (3, 64, 8)]:
+        x = (torch.randn(N, C, H, H) * 2).cuda().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last) \
+            .requires_grad_(True)
+        y = head_avgpool(x)
+        ref = torch.nn.functional.adaptive_avg_pool2d(x.float(), (1, 1))
+        assert y.shape == (N, C, 1, 1)
+        assert (y.float() - ref).abs().max() < 2e-2
+        g = torch.randn_like(y)
+        y.backward(g)
+        dref = (g.float() / (H * H)).expand(N, C, H, H)
+        assert (x.grad.float() - dref).abs().max() < 2e-2
