@@ -200,6 +200,12 @@ def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args,
 
     stepper.model.train()
     total_images = 0
+    # meter reads (.item()) force a device sync; under the hipGraph
+    # stepper that would serialize every replay, so accumulate the window
+    # sums ON DEVICE and sync only at the print cadence.  Meter `val`
+    # becomes the window mean rather than the single last batch — the
+    # running averages are identical to the reference's.
+    win = {"loss": None, "acc1": None, "acc5": None, "n": 0, "bs": 0}
     end = time.time()
     for i, (raw, _) in enumerate(train_loader):
         data_time.update(time.time() - end)
@@ -207,15 +213,28 @@ def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args,
         loss, output, target = stepper.step_from_raw(raw)
 
         acc1, acc5 = accuracy(output.detach(), target, topk=(1, 5))
-        losses.update(loss.item(), raw.size(0))
-        top1.update(acc1[0].item(), raw.size(0))
-        top5.update(acc5[0].item(), raw.size(0))
+        vals = (loss.detach().float(), acc1[0].detach(), acc5[0].detach())
+        if win["n"] == 0:
+            win["loss"], win["acc1"], win["acc5"] = \
+                [v.clone() for v in vals]
+        else:
+            win["loss"] += vals[0]
+            win["acc1"] += vals[1]
+            win["acc5"] += vals[2]
+        win["n"] += 1
+        win["bs"] = raw.size(0)
 
         batch_time.update(time.time() - end)
         end = time.time()
         total_images += raw.size(0)
 
         if i % args.print_freq == 0:
+            n = win["n"]
+            bs = raw.size(0)
+            losses.update(win["loss"].item() / n, bs * n)
+            top1.update(win["acc1"].item() / n, bs * n)
+            top5.update(win["acc5"].item() / n, bs * n)
+            win["n"] = 0
             progress.display(i)
             if metrics is not None:
                 metrics.log(epoch=epoch, step=i, loss=round(losses.val, 5),
@@ -223,6 +242,11 @@ def train_epoch(train_loader, stepper: MoCoStepper, epoch: int, args,
                             img_s=round(raw.size(0) /
                                         max(batch_time.val, 1e-9), 2),
                             data_s=round(data_time.val, 4))
+    if win["n"]:
+        n = win["n"]
+        losses.update(win["loss"].item() / n, win["bs"] * n)
+        top1.update(win["acc1"].item() / n, win["bs"] * n)
+        top5.update(win["acc5"].item() / n, win["bs"] * n)
     if metrics is not None:
         metrics.log(epoch=epoch, summary=True, loss_avg=round(losses.avg, 5),
                     acc1_avg=round(top1.avg, 3),
