@@ -143,3 +143,28 @@ def test_shuffle_unshuffle_roundtrip_world1():
     restored = model._batch_unshuffle(shuffled, idx_unshuffle)
     assert torch.equal(restored, x)
     assert not torch.equal(shuffled, x)  # permutation actually shuffles
+
+
+def test_graph_safe_deferred_enqueue_cpu():
+    """graph_safe mode: forward defers the enqueue past backward (no queue
+    snapshot); flush_queue() applies it.  Semantics on CPU match the eager
+    path exactly."""
+    m1 = make_moco()
+    m2 = make_moco()
+    m2.load_state_dict(m1.state_dict())
+    m2.graph_safe = True
+    x_q = torch.randn(8, 3, 32, 32)
+    x_k = torch.randn(8, 3, 32, 32)
+
+    q_before = m2.queue.clone()
+    logits1, _ = m1(x_q.clone(), x_k.clone(), step=0)
+    logits2, _ = m2(x_q.clone(), x_k.clone(), step=0)
+    # deferred: queue untouched until flush
+    assert torch.equal(m2.queue, q_before)
+    assert m2._pending_keys is not None
+    m2.flush_queue()
+    assert m2._pending_keys is None
+    assert torch.equal(m1.queue, m2.queue)
+    assert m1.queue_ptr.item() == m2.queue_ptr.item()
+    # world-1 logits identical (shuffle skip is exactly inverted anyway)
+    assert torch.allclose(logits1, logits2, atol=1e-5)

@@ -126,3 +126,20 @@ def test_contrastive_odd_dim_cpu_path():
     assert logits.shape == (4, 17)
     logits.sum().backward()
     assert q.grad is not None
+
+
+def test_fp8_quantize_roundtrip_cpu():
+    """Per-tensor e4m3 quantization: dequantized values track the source
+    within e4m3's relative precision, amax maps near the format max."""
+    import torch
+
+    from moco_amd.ops.fp8 import E4M3_MAX, quantize_e4m3
+    torch.manual_seed(3)
+    x = torch.randn(64, 128) * 5
+    x8, s = quantize_e4m3(x)
+    xd = x8.float() * s
+    rel = (xd - x).abs() / x.abs().clamp_min(1e-3)
+    assert float(rel.median()) < 0.04
+    assert float(x8.float().abs().max()) <= E4M3_MAX
+    # amax element lands at the top of the representable range
+    assert float(x8.float().abs().max()) > 0.9 * E4M3_MAX
