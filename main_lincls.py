@@ -243,7 +243,7 @@ def validate(loader, model, criterion, device, args):
     for i, (raw, target) in enumerate(loader):
         if args.max_steps_per_epoch and i >= args.max_steps_per_epoch:
             break
-        images = torch.stack([tfm(x) for x in raw]).to(device)
+        images = tfm(raw).to(device)  # batched resize/crop/normalize
         target = target.to(device)
         output = model(images)
         loss = criterion(output, target)

@@ -147,21 +147,26 @@ def build_dataset(data: str, load_size: int = 256, synthetic_len: int = 2048,
 
 def cpu_two_crop_eval_transform(out_size: int = 224):
     """Deterministic center-crop + normalize for eval paths (CPU tensor in,
-    CPU tensor out)."""
+    CPU tensor out).  Accepts a single [3,H,W] image or a whole [B,3,H,W]
+    batch (one interpolate/crop/normalize for the batch — no per-image
+    Python loop)."""
     mean = torch.tensor(IMAGENET_MEAN).view(3, 1, 1)
     std = torch.tensor(IMAGENET_STD).view(3, 1, 1)
 
     def _t(x: torch.Tensor) -> torch.Tensor:
-        _, h, w = x.shape
+        single = x.dim() == 3
+        if single:
+            x = x.unsqueeze(0)
+        h, w = x.shape[2], x.shape[3]
         scale = out_size / min(h, w)
         nh, nw = max(out_size, int(round(h * scale))), max(
             out_size, int(round(w * scale)))
         x = torch.nn.functional.interpolate(
-            x.unsqueeze(0), size=(nh, nw), mode="bilinear",
-            align_corners=False).squeeze(0)
+            x, size=(nh, nw), mode="bilinear", align_corners=False)
         top = (nh - out_size) // 2
         left = (nw - out_size) // 2
-        x = x[:, top:top + out_size, left:left + out_size]
-        return (x - mean) / std
+        x = x[:, :, top:top + out_size, left:left + out_size]
+        x = (x - mean) / std
+        return x.squeeze(0) if single else x
 
     return _t
