@@ -1,12 +1,10 @@
-"""1x1 stride-1 convolution on the hand-written MFMA NT-GEMM.
+"""Standalone 1x1/s1 conv on the MFMA NT-GEMM — legacy/env-gated.
 
-In channels_last, a 1x1/s1 conv collapses to out[M, Co] = x[M, Ci] @ W^T
-with M = N*H*W; its data-gradient is the same GEMM against the transposed
-weight.  Forward and dgrad run on ops/hip/conv1x1.hip; the weight gradient
-(a TN GEMM over millions of rows) stays on hipBLASLt via torch.matmul.
-
-Enabled with MOCO_AMD_CUSTOM_CONV1X1=1 (off by default until it beats the
-MIOpen/CK pick on the target shapes — see tools/conv1x1_probe.py).
+The PRODUCTION path is ops/block.py (the whole Bottleneck as one fused
+Function: same GEMM kernels plus BN prologue/epilogue fusion, MIOpen
+wgrad).  This standalone wrapper survives for A/B probing of the raw GEMM
+(tools/conv1x1_probe.py) on the generic per-op route; enable with
+MOCO_AMD_CUSTOM_CONV1X1=1.
 """
 
 import os
