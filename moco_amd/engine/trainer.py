@@ -151,9 +151,21 @@ class MoCoStepper:
                 self._gwarm += 1
                 return out
             self.optimizer.zero_grad(set_to_none=True)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g, stream=self._gstream):
-                self._gout = self._inner_step(self._gq, self._gk)
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g, stream=self._gstream):
+                    self._gout = self._inner_step(self._gq, self._gk)
+            except Exception as e:  # capture executes nothing: state is
+                # untouched on device — fall back to eager permanently
+                # rather than killing the run
+                import warnings
+                warnings.warn(f"hipGraph capture failed ({e!r}); "
+                              "continuing eager")
+                self.use_graph = False
+                raw_model._pending_keys = None
+                raw_model.graph_safe = False
+                torch.cuda.synchronize()
+                return self.step(im_q, im_k)
             self._graph = g
             # capture itself executes nothing: replay once to take the step
             g.replay()

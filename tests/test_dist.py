@@ -176,14 +176,16 @@ def _worker_bench_codepath(rank, world, port):
     # queue replicated bit-exactly across ranks
     qs = [torch.empty_like(raw.queue) for _ in range(world)]
     dist.all_gather(qs, raw.queue)
-    assert torch.equal(qs[0], qs[1]), "queue diverged across ranks"
+    for r in range(1, world):
+        assert torch.equal(qs[0], qs[r]), "queue diverged across ranks"
     assert raw.queue_ptr.item() == (4 * 4 * world) % 64
     # params identical across ranks after DDP-synced updates
     flat = torch.cat([p.detach().reshape(-1)
                       for p in raw.parameters()])[:100000]
     ps = [torch.empty_like(flat) for _ in range(world)]
     dist.all_gather(ps, flat)
-    assert torch.equal(ps[0], ps[1]), "params diverged across ranks"
+    for r in range(1, world):
+        assert torch.equal(ps[0], ps[r]), "params diverged across ranks"
     dist.destroy_process_group()
 
 
@@ -225,3 +227,7 @@ def _worker_a2a_plan(rank, world, port):
 
 def test_a2a_plan_matches_bruteforce():
     _run_spawn(_worker_a2a_plan, 29319)
+
+
+def test_bench_codepath_world4_numerics():
+    _run_spawn(_worker_bench_codepath, 29320, world=4)
