@@ -143,3 +143,18 @@ def test_fp8_quantize_roundtrip_cpu():
     assert float(x8.float().abs().max()) <= E4M3_MAX
     # amax element lands at the top of the representable range
     assert float(x8.float().abs().max()) > 0.9 * E4M3_MAX
+
+
+def test_eval_transform_batched_matches_single():
+    """cpu_two_crop_eval_transform on a [B,3,H,W] batch must equal the
+    per-image application (the lincls validate path batches it)."""
+    import torch
+
+    from moco_amd.loader import cpu_two_crop_eval_transform
+    torch.manual_seed(5)
+    tfm = cpu_two_crop_eval_transform(32)
+    batch = torch.rand(5, 3, 48, 40)
+    out_b = tfm(batch)
+    out_s = torch.stack([tfm(batch[i]) for i in range(5)])
+    assert out_b.shape == (5, 3, 32, 32)
+    assert torch.allclose(out_b, out_s, atol=1e-6)
