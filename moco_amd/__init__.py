@@ -6,7 +6,7 @@ Built from scratch with the capabilities of the reference MoCo codebase
 (see SURVEY.md); not a port.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from . import builder, loader, models, ops, parallel, utils  # noqa: F401
 from .builder import MoCo, moco_loss  # noqa: F401
