@@ -158,3 +158,56 @@ def test_eval_transform_batched_matches_single():
     out_s = torch.stack([tfm(batch[i]) for i in range(5)])
     assert out_b.shape == (5, 3, 32, 32)
     assert torch.allclose(out_b, out_s, atol=1e-6)
+
+
+def test_mirror_set_table_construction():
+    """MirrorSet chunk/tile tables (CPU-constructible): every Bottleneck
+    1x1 conv gets a [Co,Ci] bf16 mirror + [Ci,Co] transposed mirror, the
+    3x3 a channels_last mirror; cast chunks tile the full numel and the
+    transpose meta covers every 64x64 tile exactly once."""
+    import torch
+
+    from moco_amd.models.resnet import resnet50
+    from moco_amd.ops.mirrors import MirrorSet
+
+    m = resnet50(num_classes=8)
+    ms = MirrorSet(m)
+    # 16 bottlenecks x 3 convs mirrored
+    assert len(ms.weights) == 48
+    assert len(ms.transposed) == 32  # conv1 + conv3 per block
+    for w, mb in ms.weights:
+        assert mb.dtype == torch.bfloat16
+        assert mb.numel() == w.numel()
+    # cast chunks cover every element exactly once
+    total = sum(int(s) for s in ms._cast_sz)
+    assert total == sum(w.numel() for w, _ in ms.weights)
+    # transpose tiles: sum of 64x64 tiles == transposed elements
+    assert ms._t_meta.numel() == 4 * ms._t_sp.numel()
+    tiles = ms._t_sp.numel()
+    assert tiles * 64 * 64 == sum(co * ci for _, _, co, ci in ms.transposed)
+    for mb, wt, co, ci in ms.transposed:
+        assert wt.shape == (ci, co) and co % 64 == 0 and ci % 64 == 0
+    assert ms.valid()
+
+
+def test_kernel_buckets_tool():
+    """The profiling summarizer parses a rocprofv3-style stats CSV."""
+    import io
+    import subprocess
+    import sys
+
+    csv = ('"Name","Calls","TotalDurationNs","AverageNs","Percentage",'
+           '"MinNs","MaxNs","StdDev"\n'
+           '"gemm_nt_lds_kernel<128, true, true, true>",10,5000000,'
+           '500000,50.0,1,2,0\n'
+           '"naive_conv_x",2,5000000,2500000,50.0,1,2,0\n')
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".csv",
+                                     delete=False) as f:
+        f.write(csv)
+        path = f.name
+    r = subprocess.run([sys.executable, "tools/kernel_buckets.py", path,
+                        "2"], capture_output=True, text=True, timeout=60)
+    assert r.returncode == 0, r.stderr
+    assert "our gemm_nt fwd" in r.stdout
+    assert "miopen naive_conv" in r.stdout
